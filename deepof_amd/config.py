@@ -1,0 +1,104 @@
+"""Single dataclass config with YAML + CLI overrides.
+
+Replaces the reference's scattered tf.app.flags / hardcoded
+hyper-params; every field of the §2.5 config matrix is a
+named field here.
+"""
+
+from __future__ import annotations
+
+import dataclasses
+from dataclasses import dataclass, field
+from typing import Optional
+
+import yaml
+
+
+@dataclass
+class Config:
+    # experiment
+    run_name: str = "run"
+    log_dir: str = "logs"
+    seed: int = 0
+
+    # data
+    dataset: str = "synthetic"           # synthetic|flying_chairs|sintel|ucf101
+    data_dir: str = ""
+    image_size: tuple = (384, 512)       # (H, W)
+    crop_size: Optional[tuple] = None
+    batch_size: int = 8
+    num_workers: int = 4
+    time_step: int = 2                   # >2 = Sintel multi-frame volumes
+    sintel_pass: str = "clean"
+
+    # model
+    model: str = "flownets"              # flownets|flownetc|vgg16|inception_v3
+    activation: str = "elu"
+
+    # loss (SURVEY §2.5 defaults: chairs FlowNetS v0)
+    epsilon: float = 1e-4
+    alpha_c: float = 0.25
+    alpha_s: float = 0.37
+    lambda_smooth: float = 1.0
+    loss_weights: Optional[list] = None  # default from model registry
+    guided: bool = False                 # add proxy-label supervision
+    guided_weight: float = 1.0
+    photo_weight: float = 1.0
+
+    # optimization
+    lr: float = 1.6e-5
+    lr_decay: float = 0.5
+    epochs_per_decay: int = 18
+    adam_beta1: float = 0.9
+    adam_beta2: float = 0.999
+    adam_eps: float = 1e-8
+    weight_decay: float = 0.0
+    max_epochs: int = 110
+    grad_clip: float = 0.0
+
+    # runtime
+    precision: str = "bf16"              # bf16|fp32
+    device: str = "cuda"
+    channels_last: bool = False
+    save_interval_epochs: int = 5
+    log_interval: int = 50
+    eval_interval_epochs: int = 1
+    resume: bool = True
+
+    # action head (UCF101 joint training)
+    action_classes: int = 0              # >0 enables the action head
+    action_weight: float = 1.0
+
+    def to_dict(self):
+        return dataclasses.asdict(self)
+
+    @classmethod
+    def from_yaml(cls, path: str) -> "Config":
+        with open(path) as f:
+            data = yaml.safe_load(f) or {}
+        return cls.from_dict(data)
+
+    @classmethod
+    def from_dict(cls, data: dict) -> "Config":
+        names = {f.name for f in dataclasses.fields(cls)}
+        unknown = set(data) - names
+        if unknown:
+            raise ValueError(f"unknown config keys: {sorted(unknown)}")
+        cfg = cls(**data)
+        if isinstance(cfg.image_size, list):
+            cfg.image_size = tuple(cfg.image_size)
+        if isinstance(cfg.crop_size, list):
+            cfg.crop_size = tuple(cfg.crop_size)
+        return cfg
+
+    def apply_overrides(self, overrides: list[str]) -> "Config":
+        """key=value overrides (YAML-parsed values)."""
+        data = self.to_dict()
+        for ov in overrides:
+            if "=" not in ov:
+                raise ValueError(f"override must be key=value, got {ov!r}")
+            k, v = ov.split("=", 1)
+            if k not in data:
+                raise ValueError(f"unknown config key {k!r}")
+            data[k] = yaml.safe_load(v)
+        return Config.from_dict(data)

@@ -1,0 +1,61 @@
+"""Evaluation: the reference AEE protocol.
+
+Parity (/root/reference/flyingChairsTrain.py:264-296): the final flow is
+pr1 (finest prediction, at HALF input resolution) times its flow scale,
+times 2 ("pr1 is still half of the final predicted flow value"), clipped
+to the dataset's training flow range, bilinearly resized to the
+ground-truth resolution WITHOUT magnitude rescaling (a reference quirk,
+preserved for comparability), then AEE = mean endpoint error.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from .. import ops
+from ..losses.unsup import preprocess_images
+
+# dataset -> (multiplier, clip_min, clip_max) — SURVEY §2.5 eval row
+EVAL_POSTPROC = {
+    "flying_chairs": (2.0, -300.0, 250.0),
+    "synthetic": (2.0, -300.0, 250.0),
+    "sintel": (3.0, -420.621, 426.311),
+    "ucf101": (2.0, -300.0, 250.0),
+}
+
+
+@torch.no_grad()
+def predict_flow(model, img1_raw, img2_raw, mean_bgr, flow_scale_finest,
+                 dataset: str = "flying_chairs",
+                 gt_size: tuple[int, int] | None = None) -> torch.Tensor:
+    """Run the model and apply the eval post-processing; returns [B,2,H,W]."""
+    x1 = preprocess_images(img1_raw.float(), mean_bgr)
+    x2 = preprocess_images(img2_raw.float(), mean_bgr)
+    flows = model(torch.cat([x1, x2], dim=1))
+    mult, cmin, cmax = EVAL_POSTPROC.get(dataset, EVAL_POSTPROC["flying_chairs"])
+    pred = flows[0].float() * flow_scale_finest * mult
+    pred = pred.clamp(cmin, cmax)
+    if gt_size is not None and tuple(pred.shape[-2:]) != tuple(gt_size):
+        # bilinear resize with NO magnitude rescale (reference quirk)
+        pred = ops.resize_bilinear(pred, gt_size[0], gt_size[1])
+    return pred
+
+
+@torch.no_grad()
+def evaluate_aee(model, loader, mean_bgr, flow_scale_finest, device,
+                 dataset: str = "flying_chairs", max_batches=None) -> float:
+    model.eval()
+    total = 0.0
+    count = 0
+    for i, batch in enumerate(loader):
+        if max_batches is not None and i >= max_batches:
+            break
+        img1 = batch["img1"].to(device, non_blocking=True)
+        img2 = batch["img2"].to(device, non_blocking=True)
+        gt = batch["flow"].to(device, non_blocking=True).float()
+        pred = predict_flow(model, img1, img2, mean_bgr, flow_scale_finest,
+                            dataset, gt_size=tuple(gt.shape[-2:]))
+        total += float(ops.endpoint_error_sum(pred, gt))
+        count += gt.shape[0] * gt.shape[-2] * gt.shape[-1]
+    model.train()
+    return total / max(count, 1)

@@ -1,0 +1,7 @@
+from .flownet import FlowNetS, FlowNetC
+from .vgg16 import VGG16Flow
+from .inception import InceptionFlow
+from .registry import build_model, MODEL_REGISTRY
+
+__all__ = ["FlowNetS", "FlowNetC", "VGG16Flow", "InceptionFlow",
+           "build_model", "MODEL_REGISTRY"]
