@@ -1,0 +1,143 @@
+"""Flagship benchmark: FlowNetS unsupervised training on FlyingChairs-
+shaped synthetic data, 512x384 bf16, batch 64/GPU (BASELINE.json
+configs[1]), 1 process per GPU over RCCL.
+
+    python bench.py --gpus 1 --steps 20 --warmup 5
+    torchrun --nproc-per-node 8 bench.py --gpus 8 ...
+
+Prints ONE JSON line from rank 0: whole-job imgs/sec (max step time
+over ranks), weak scaling.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=64)
+    ap.add_argument("--height", type=int, default=384)
+    ap.add_argument("--width", type=int, default=512)
+    ap.add_argument("--model", type=str, default="flownets")
+    ap.add_argument("--dtype", type=str, default="bf16")
+    args = ap.parse_args()
+
+    import torch.distributed as dist
+
+    from deepof_amd.engine.optim import FusedAdam
+    from deepof_amd.losses import MultiScaleUnsupLoss, preprocess_images
+    from deepof_amd.losses.unsup import DATASET_MEANS
+    from deepof_amd.models import build_model
+    from deepof_amd.parallel import BucketedDataParallel
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if world > 1:
+        dist.init_process_group("nccl")
+    torch.cuda.set_device(local_rank)
+    dev = torch.device(f"cuda:{local_rank}")
+    torch.manual_seed(1234 + rank)
+
+    model, flow_scales, weights = build_model(args.model)
+    model.to(dev)
+    mean = DATASET_MEANS["flying_chairs"]
+    loss_fn = MultiScaleUnsupLoss(flow_scales, weights, mean)
+    if world > 1:
+        model = BucketedDataParallel(model)
+    opt = FusedAdam(model.parameters(), lr=1.6e-5)
+
+    # synthetic FlyingChairs-shaped data, resident on device; a few
+    # distinct batches cycle so no step sees cached activations
+    n_batches = 4
+    batches = []
+    for i in range(n_batches):
+        img1 = torch.rand(args.batch, 3, args.height, args.width,
+                          device=dev) * 255
+        img2 = img1.roll(shifts=(2, -3), dims=(2, 3)) * 0.9 + \
+            torch.rand_like(img1) * 0.1 * 255
+        x = torch.cat([preprocess_images(img1, mean),
+                       preprocess_images(img2, mean)], dim=1)
+        batches.append((x, img1, img2))
+
+    use_bf16 = args.dtype == "bf16"
+
+    def step(i):
+        x, img1, img2 = batches[i % n_batches]
+        with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16):
+            flows = model(x)
+        res = loss_fn(flows, img1, img2)
+        res["total"].backward()
+        if isinstance(model, BucketedDataParallel):
+            model.finish_gradient_sync()
+        opt.step()
+        if isinstance(model, BucketedDataParallel):
+            model.zero_grad_buckets()
+        else:
+            opt.zero_grad(set_to_none=False)
+        return res["total"]
+
+    for i in range(args.warmup):
+        step(i)
+
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    last = None
+    for i in range(args.steps):
+        last = step(i)
+    torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    if world > 1:
+        t = torch.tensor([elapsed], device=dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t)
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    imgs_per_sec = args.batch * world * args.steps / elapsed
+    if rank == 0:
+        print(json.dumps({
+            "metric": "imgs/sec FlowNetS@FlyingChairs",
+            "value": imgs_per_sec,
+            "unit": "imgs/sec",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "loss_last": float(last.detach()),
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * world,
+                "seq_len": None,
+                "image_size": [args.height, args.width],
+                "parallelism": f"dp{world}",
+            },
+        }))
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

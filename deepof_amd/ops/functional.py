@@ -184,9 +184,14 @@ def unsup_loss_scale(
     b, c, h, w = img1.shape
     import math
 
+    # mirror the kernel's border-mask fallback for degenerate scales
     bw = math.ceil(h * 0.1)
-    num_valid = float(b * c * (h - 2 * bw) * (w - 2 * bw))
-    num_valid_flows = num_valid / c * 2
+    if (h - 2 * bw) > 0 and (w - 2 * bw) > 0:
+        num_valid = float(b * c * (h - 2 * bw) * (w - 2 * bw))
+        num_valid_flows = num_valid / c * 2
+    else:
+        num_valid = float(b * c * h * w)
+        num_valid_flows = float(2 * b * h * w)
     photo_sum, u_sum, v_sum, recon = _FusedUnsupLoss.apply(
         flow_raw.float(), img1, img2, flow_scale, epsilon, alpha_c, alpha_s,
         return_recon,

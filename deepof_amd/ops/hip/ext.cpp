@@ -1,0 +1,42 @@
+// Python bindings for the deepof_amd gfx950 HIP extension.
+
+#include <torch/extension.h>
+
+at::Tensor warp_forward(at::Tensor img2, at::Tensor flow);
+std::vector<at::Tensor> warp_backward(at::Tensor grad_out, at::Tensor img2,
+                                      at::Tensor flow);
+std::vector<at::Tensor> unsup_loss_forward(at::Tensor flow, at::Tensor img1,
+                                           at::Tensor img2, double scale,
+                                           double eps, double alpha_c,
+                                           double alpha_s, bool want_recon);
+at::Tensor unsup_loss_backward(at::Tensor flow, at::Tensor img1,
+                               at::Tensor img2, double scale, double eps,
+                               double alpha_c, double alpha_s, double g_photo,
+                               double g_u, double g_v);
+at::Tensor resize_bilinear(at::Tensor x, long oh, long ow);
+at::Tensor lrn_forward(at::Tensor x, long radius, double bias, double alpha,
+                       double beta);
+at::Tensor epe_sum(at::Tensor f, at::Tensor g);
+at::Tensor correlation_forward(at::Tensor f1, at::Tensor f2, long md);
+std::vector<at::Tensor> correlation_backward(at::Tensor gout, at::Tensor f1,
+                                             at::Tensor f2, long md);
+void fused_adam(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+                std::vector<at::Tensor> exp_avgs,
+                std::vector<at::Tensor> exp_avg_sqs, double lr, double beta1,
+                double beta2, double eps, double wd, double bias1,
+                double bias2);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("warp_forward", &warp_forward, "bilinear backward warp (fwd)");
+  m.def("warp_backward", &warp_backward, "bilinear backward warp (bwd)");
+  m.def("unsup_loss_forward", &unsup_loss_forward,
+        "fused warp+Charbonnier+smoothness (fwd sums)");
+  m.def("unsup_loss_backward", &unsup_loss_backward,
+        "fused loss backward -> d(flow)");
+  m.def("resize_bilinear", &resize_bilinear, "legacy-TF bilinear resize");
+  m.def("lrn_forward", &lrn_forward, "across-channel LRN");
+  m.def("epe_sum", &epe_sum, "endpoint-error sum reduction");
+  m.def("correlation_forward", &correlation_forward, "cost volume fwd");
+  m.def("correlation_backward", &correlation_backward, "cost volume bwd");
+  m.def("fused_adam", &fused_adam, "multi-tensor Adam step");
+}

@@ -1,0 +1,225 @@
+"""GPU numerics: every HIP kernel vs the pure-torch fp32 reference."""
+
+import pytest
+import torch
+
+from deepof_amd.ops import reference as ref
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def _hip():
+    from deepof_amd.ops.functional import require_hip
+
+    return require_hip()
+
+
+def test_extension_loaded():
+    from deepof_amd.ops.functional import hip_available
+
+    assert hip_available(), "HIP extension must be built on a GPU box"
+
+
+def test_warp_forward_matches_reference():
+    torch.manual_seed(0)
+    img2 = torch.randn(2, 3, 37, 53)
+    flow = torch.randn(2, 2, 37, 53) * 5
+    want = ref.warp_bilinear(img2, flow)
+    got = _hip().warp_forward(img2.to(DEV), flow.to(DEV)).cpu()
+    torch.testing.assert_close(got, want, rtol=1e-5, atol=1e-5)
+
+
+def test_warp_forward_bf16():
+    torch.manual_seed(0)
+    img2 = torch.randn(1, 3, 32, 48)
+    flow = torch.randn(1, 2, 32, 48) * 3
+    want = ref.warp_bilinear(img2, flow)
+    got = _hip().warp_forward(img2.to(DEV).bfloat16(), flow.to(DEV)).float().cpu()
+    torch.testing.assert_close(got, want, rtol=0.05, atol=0.05)
+
+
+def test_warp_backward_matches_autograd():
+    torch.manual_seed(0)
+    img2 = torch.randn(1, 3, 24, 30, requires_grad=True)
+    flow = (torch.randn(1, 2, 24, 30) * 2).requires_grad_(True)
+    out = ref.warp_bilinear(img2, flow)
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    gi, gf = _hip().warp_backward(g.to(DEV), img2.detach().to(DEV),
+                                  flow.detach().to(DEV))
+    torch.testing.assert_close(gf.cpu(), flow.grad, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(gi.cpu(), img2.grad, rtol=1e-4, atol=1e-4)
+
+
+@pytest.mark.parametrize("hw", [(64, 96), (30, 40)])
+def test_unsup_loss_forward_matches_reference(hw):
+    torch.manual_seed(0)
+    h, w = hw
+    flow = torch.randn(2, 2, h, w)
+    img1 = torch.rand(2, 3, h, w)
+    img2 = torch.rand(2, 3, h, w)
+    want = ref.unsup_loss_scale(flow, img1, img2, flow_scale=2.5,
+                                lambda_smooth=1.0)
+
+    from deepof_amd.ops import unsup_loss_scale
+
+    got = unsup_loss_scale(flow.to(DEV), img1.to(DEV), img2.to(DEV),
+                           flow_scale=2.5, lambda_smooth=1.0)
+    for k in ("total", "photo", "u_loss", "v_loss"):
+        torch.testing.assert_close(got[k].cpu(), want[k],
+                                   rtol=1e-4, atol=1e-5)
+
+
+def test_unsup_loss_backward_matches_autograd():
+    torch.manual_seed(0)
+    h, w = 40, 56
+    flow_cpu = torch.randn(1, 2, h, w, requires_grad=True)
+    img1 = torch.rand(1, 3, h, w)
+    img2 = torch.rand(1, 3, h, w)
+    want = ref.unsup_loss_scale(flow_cpu, img1, img2, flow_scale=5.0)
+    want["total"].backward()
+
+    from deepof_amd.ops import unsup_loss_scale
+
+    flow_gpu = flow_cpu.detach().to(DEV).requires_grad_(True)
+    got = unsup_loss_scale(flow_gpu, img1.to(DEV), img2.to(DEV),
+                           flow_scale=5.0)
+    got["total"].backward()
+    torch.testing.assert_close(flow_gpu.grad.cpu(), flow_cpu.grad,
+                               rtol=2e-4, atol=2e-5)
+
+
+def test_unsup_loss_recon():
+    torch.manual_seed(0)
+    flow = torch.randn(1, 2, 32, 32)
+    img1 = torch.rand(1, 3, 32, 32)
+    img2 = torch.rand(1, 3, 32, 32)
+    from deepof_amd.ops import unsup_loss_scale
+
+    got = unsup_loss_scale(flow.to(DEV), img1.to(DEV), img2.to(DEV),
+                           flow_scale=1.0, return_recon=True)
+    want = ref.warp_bilinear(img2, flow)
+    torch.testing.assert_close(got["recon"].cpu(), want, rtol=1e-5, atol=1e-5)
+
+
+def test_resize_bilinear_matches_reference():
+    torch.manual_seed(0)
+    x = torch.randn(2, 3, 37, 53)
+    for oh, ow in [(19, 27), (74, 106), (37, 53)]:
+        want = ref.resize_bilinear(x, oh, ow)
+        got = _hip().resize_bilinear(x.to(DEV), oh, ow).cpu()
+        torch.testing.assert_close(got, want, rtol=1e-5, atol=1e-5)
+
+
+def test_lrn_matches_reference():
+    torch.manual_seed(0)
+    x = torch.randn(2, 3, 17, 23)
+    want = ref.lrn(x)
+    got = _hip().lrn_forward(x.to(DEV), 4, 1.0, 1.0, 0.7).cpu()
+    torch.testing.assert_close(got, want, rtol=1e-5, atol=1e-5)
+
+
+def test_epe_sum():
+    torch.manual_seed(0)
+    f = torch.randn(2, 2, 31, 41)
+    g = torch.randn(2, 2, 31, 41)
+    want = torch.sqrt(((f - g) ** 2).sum(dim=1)).sum()
+    got = _hip().epe_sum(f.to(DEV), g.to(DEV)).cpu()
+    torch.testing.assert_close(got, want, rtol=1e-5, atol=1e-3)
+
+
+def test_correlation_matches_reference():
+    torch.manual_seed(0)
+    f1 = torch.randn(2, 16, 20, 28)
+    f2 = torch.randn(2, 16, 20, 28)
+    md = 3
+    want = ref.correlation(f1, f2, md)
+    got = _hip().correlation_forward(f1.to(DEV), f2.to(DEV), md).cpu()
+    torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-5)
+
+
+def test_correlation_backward_matches_autograd():
+    torch.manual_seed(0)
+    f1 = torch.randn(1, 8, 10, 12, requires_grad=True)
+    f2 = torch.randn(1, 8, 10, 12, requires_grad=True)
+    md = 2
+    out = ref.correlation(f1, f2, md)
+    g = torch.randn_like(out)
+    out.backward(g)
+    g1, g2 = _hip().correlation_backward(g.to(DEV), f1.detach().to(DEV),
+                                         f2.detach().to(DEV), md)
+    torch.testing.assert_close(g1.cpu(), f1.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(g2.cpu(), f2.grad, rtol=1e-4, atol=1e-5)
+
+
+def test_fused_adam_matches_cpu():
+    from deepof_amd.engine.optim import FusedAdam
+
+    torch.manual_seed(0)
+    shapes = [(100,), (33, 7), (5, 3, 3, 3), (1025,)]
+    cpu_params = [torch.randn(s, requires_grad=True) for s in shapes]
+    gpu_params = [p.detach().clone().to(DEV).requires_grad_(True)
+                  for p in cpu_params]
+    grads = [torch.randn(s) for s in shapes]
+    for p, g in zip(cpu_params, grads):
+        p.grad = g.clone()
+    for p, g in zip(gpu_params, grads):
+        p.grad = g.to(DEV)
+
+    opt_c = FusedAdam(cpu_params, lr=1e-2, weight_decay=0.01)
+    opt_g = FusedAdam(gpu_params, lr=1e-2, weight_decay=0.01)
+    for _ in range(3):
+        opt_c.step()
+        opt_g.step()
+    for pc, pg in zip(cpu_params, gpu_params):
+        torch.testing.assert_close(pg.detach().cpu(), pc.detach(),
+                                   rtol=1e-5, atol=1e-6)
+
+
+def test_model_step_gpu():
+    """Full FlowNetS fwd+bwd+step in bf16 on GPU produces finite loss."""
+    from deepof_amd.engine.optim import FusedAdam
+    from deepof_amd.losses import MultiScaleUnsupLoss, preprocess_images
+    from deepof_amd.models import build_model
+
+    torch.manual_seed(0)
+    model, scales, weights = build_model("flownets")
+    model.to(DEV)
+    loss_fn = MultiScaleUnsupLoss(scales, weights)
+    opt = FusedAdam(model.parameters(), lr=1e-5)
+    img1 = torch.rand(2, 3, 192, 256, device=DEV) * 255
+    img2 = torch.rand(2, 3, 192, 256, device=DEV) * 255
+    x = torch.cat([preprocess_images(img1, loss_fn.mean_bgr),
+                   preprocess_images(img2, loss_fn.mean_bgr)], dim=1)
+    losses = []
+    for _ in range(3):
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            flows = model(x)
+        res = loss_fn(flows, img1, img2)
+        res["total"].backward()
+        opt.step()
+        opt.zero_grad(set_to_none=False)
+        losses.append(float(res["total"]))
+    assert all(l == l for l in losses), losses  # no NaN
+
+
+def test_flownetc_step_gpu():
+    from deepof_amd.losses import MultiScaleUnsupLoss, preprocess_images
+    from deepof_amd.models import build_model
+
+    torch.manual_seed(0)
+    model, scales, weights = build_model("flownetc")
+    model.to(DEV)
+    loss_fn = MultiScaleUnsupLoss(scales, weights)
+    img1 = torch.rand(1, 3, 192, 256, device=DEV) * 255
+    img2 = torch.rand(1, 3, 192, 256, device=DEV) * 255
+    x = torch.cat([preprocess_images(img1, loss_fn.mean_bgr),
+                   preprocess_images(img2, loss_fn.mean_bgr)], dim=1)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        flows = model(x)
+    res = loss_fn(flows, img1, img2)
+    res["total"].backward()
+    assert float(res["total"]) == float(res["total"])
