@@ -32,6 +32,9 @@ def main():
     ap.add_argument("--width", type=int, default=512)
     ap.add_argument("--model", type=str, default="flownets")
     ap.add_argument("--dtype", type=str, default="bf16")
+    ap.add_argument("--channels-last", action="store_true", default=True)
+    ap.add_argument("--no-channels-last", dest="channels_last",
+                    action="store_false")
     args = ap.parse_args()
 
     import torch.distributed as dist
@@ -51,8 +54,11 @@ def main():
     dev = torch.device(f"cuda:{local_rank}")
     torch.manual_seed(1234 + rank)
 
+    torch.backends.cudnn.benchmark = True  # MIOpen find for each conv shape
     model, flow_scales, weights = build_model(args.model)
     model.to(dev)
+    if args.channels_last:
+        model.to(memory_format=torch.channels_last)
     mean = DATASET_MEANS["flying_chairs"]
     loss_fn = MultiScaleUnsupLoss(flow_scales, weights, mean)
     if world > 1:
@@ -70,6 +76,8 @@ def main():
             torch.rand_like(img1) * 0.1 * 255
         x = torch.cat([preprocess_images(img1, mean),
                        preprocess_images(img2, mean)], dim=1)
+        if args.channels_last:
+            x = x.to(memory_format=torch.channels_last)
         batches.append((x, img1, img2))
 
     use_bf16 = args.dtype == "bf16"
