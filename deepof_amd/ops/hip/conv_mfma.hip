@@ -1,0 +1,271 @@
+// Hand-written MFMA implicit-GEMM convolution for gfx950 (CDNA4).
+//
+// Forward conv + bias + ELU fused, NHWC bf16 activations, fp32
+// accumulate.  GEMM view: O[M=B*OH*OW, N=K] = A[M, RSC] * B[RSC, N];
+// the K-loop walks (r, s) filter taps x C-chunks, so every A sub-tile
+// load is a contiguous NHWC channel run (coalesced 16-B lane loads)
+// and B sub-tiles are contiguous rows of the [K][R][S][C] weight
+// (channels_last Conv2d layout).
+//
+// Tiling: 256 threads = 4 waves (2x2), block tile BM=128 pixels x
+// BN=128 channels, each wave a 64x64 sub-tile = 4x4 fragments of
+// v_mfma_f32_16x16x32_bf16, BK=64 reduction per stage, double-buffered
+// LDS (A 128x64 + B 128x64 bf16 = 32 KiB per stage) with the T2 XOR
+// swizzle on both tiles so ds_read_b128 is bank-conflict-free.
+//
+// Replaces (when faster — the host autotuner decides per shape) the
+// MIOpen igemm path for the FlowNetS/VGG encoder stacks
+// (/root/reference/flyingChairsWrapFlow.py:31-40) and fuses the ELU
+// that PyTorch otherwise runs as a separate elementwise pass.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include "common.hip.h"
+
+static inline hipStream_t deepof_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+namespace {
+
+using bf16 = __bf16;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) short short8v;
+
+// activation: 0 = none, 1 = ELU, 2 = LeakyReLU(0.1)
+template <int ACT>
+__device__ inline float act_fn(float x) {
+  if (ACT == 1) return x > 0.f ? x : expf(x) - 1.f;
+  if (ACT == 2) return x > 0.f ? x : 0.1f * x;
+  return x;
+}
+
+// XOR swizzle on bf16 element index within a [ROW][64] tile:
+// spread each 16-lane ds_read_b128 group across 8 16-B slots.
+__device__ inline int swz(int row, int col) {
+  return col ^ ((row & 7) << 3);
+}
+
+// ---------------------------------------------------------------------
+// fused conv+bias+act forward
+//   x:   [B, IH, IW, C]   (NHWC bf16)
+//   w:   [K, R, S, C]     (channels_last Conv2d weight, bf16)
+//   out: [B, OH, OW, K]   (NHWC bf16)
+// Grid: (ceil(M/BM) * ceil(K/BN)) blocks, 256 threads.
+// Requires C % CBK == 0 where CBK = min(C, BK).
+// ---------------------------------------------------------------------
+template <int BM, int BN, int BK, int ACT>
+__global__ __launch_bounds__(256)
+void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
+                          const bf16* __restrict__ w,
+                          const float* __restrict__ bias,
+                          bf16* __restrict__ out,
+                          int B, int IH, int IW, int C,
+                          int K, int R, int S, int OH, int OW,
+                          int stride, int pad, int n_tiles_n) {
+  __shared__ bf16 lds_a[2][BM * BK];
+  __shared__ bf16 lds_b[2][BN * BK];
+
+  const int tile_m = blockIdx.x / n_tiles_n;
+  const int tile_n = blockIdx.x % n_tiles_n;
+  const int m0 = tile_m * BM;
+  const int n0 = tile_n * BN;
+  const int M = B * OH * OW;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;                    // 4 waves
+  constexpr int WAVES_N = BN / 64;             // 2 for BN=128, 1 for BN=64
+  constexpr int WAVES_M = 4 / WAVES_N;
+  constexpr int M_FRAGS = BM / WAVES_M / 16;   // 16x16 fragments per wave
+  constexpr int N_FRAGS = 4;                   // 64 cols per wave
+  const int wm = (wid / WAVES_N) * (BM / WAVES_M);
+  const int wn = (wid % WAVES_N) * 64;
+
+  // ---- per-thread A-load geometry (BM x BK tile, 16B per lane-load) ----
+  // thread t loads rows; each row of A needs BK*2 = 128 B = 8 lane-loads.
+  // 256 threads * 16 B = 4 KiB per pass; BM*BK*2 = 16 KiB -> 4 passes.
+  // Layout: loader covers (row, kchunk): row = t / (BK/8), kchunk = t % (BK/8)
+  constexpr int KCHUNKS = BK / 8;              // 16-B chunks per row
+  constexpr int ROWS_PER_PASS = 256 / KCHUNKS; // rows loaded per pass
+  const int lrow = tid / KCHUNKS;
+  const int lk = (tid % KCHUNKS) * 8;
+
+  // precompute this thread's A pixel coords for each pass
+  int a_b[BM / ROWS_PER_PASS], a_oy[BM / ROWS_PER_PASS],
+      a_ox[BM / ROWS_PER_PASS];
+#pragma unroll
+  for (int p = 0; p < BM / ROWS_PER_PASS; ++p) {
+    const int m = m0 + lrow + p * ROWS_PER_PASS;
+    const int mm = m < M ? m : M - 1;
+    a_ox[p] = mm % OW;
+    a_oy[p] = (mm / OW) % OH;
+    a_b[p] = mm / (OW * OH);
+    if (m >= M) a_b[p] = -1;  // padded row -> zeros
+  }
+
+  f32x4 acc[M_FRAGS][N_FRAGS] = {};
+
+  const int CBK = (C < BK) ? C : BK;     // channels per k-stage chunk
+  const int taps_per_stage = BK / CBK;   // (r,s) taps packed per stage
+  const int n_stages = (R * S * C) / BK; // total K stages
+  const long ihw = (long)IH * IW;
+
+  // ---- staging helpers ----
+  auto stage = [&](int stage_idx, int buf) {
+    // reduction range [stage_idx*BK, stage_idx*BK + BK) of (r,s,c)
+    // A tile
+#pragma unroll
+    for (int p = 0; p < BM / ROWS_PER_PASS; ++p) {
+      const int rsc = stage_idx * BK + lk;  // global reduction index
+      const int c = rsc % C;
+      const int rs = rsc / C;
+      const int s = rs % S, r = rs / S;
+      bf16x8 v = {};
+      if (a_b[p] >= 0) {
+        const int iy = a_oy[p] * stride + r - pad;
+        const int ix = a_ox[p] * stride + s - pad;
+        if (iy >= 0 && iy < IH && ix >= 0 && ix < IW) {
+          const bf16* src = x + (((long)a_b[p] * IH + iy) * IW + ix) * C + c;
+          v = *reinterpret_cast<const bf16x8*>(src);
+        }
+      }
+      const int row = lrow + p * ROWS_PER_PASS;
+      *reinterpret_cast<bf16x8*>(&lds_a[buf][row * BK + swz(row, lk)]) = v;
+    }
+    // B tile: rows = output channel n, cols = reduction
+#pragma unroll
+    for (int p = 0; p < BN / ROWS_PER_PASS; ++p) {
+      const int n = n0 + lrow + p * ROWS_PER_PASS;
+      const int rsc = stage_idx * BK + lk;
+      bf16x8 v = {};
+      if (n < K) {
+        const bf16* src = w + (long)n * (R * S * C) + rsc;
+        v = *reinterpret_cast<const bf16x8*>(src);
+      }
+      const int row = lrow + p * ROWS_PER_PASS;
+      *reinterpret_cast<bf16x8*>(&lds_b[buf][row * BK + swz(row, lk)]) = v;
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+
+  for (int st = 0; st < n_stages; ++st) {
+    const int buf = st & 1;
+    if (st + 1 < n_stages) stage(st + 1, buf ^ 1);
+
+    // MFMA over this stage: BK reduction = BK/32 mfma k-steps
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 32) {
+      // load fragments: lane holds 8 contiguous reduction elems
+      // A: row = wm + mi*16 + (lane&15), k = kk + (lane>>4)*8
+      bf16x8 afrag[M_FRAGS], bfrag[N_FRAGS];
+#pragma unroll
+      for (int mi = 0; mi < M_FRAGS; ++mi) {
+        const int row = wm + mi * 16 + (lane & 15);
+        const int col = kk + (lane >> 4) * 8;
+        afrag[mi] = *reinterpret_cast<const bf16x8*>(
+            &lds_a[buf][row * BK + swz(row, col)]);
+      }
+#pragma unroll
+      for (int ni = 0; ni < N_FRAGS; ++ni) {
+        const int row = wn + ni * 16 + (lane & 15);
+        const int col = kk + (lane >> 4) * 8;
+        bfrag[ni] = *reinterpret_cast<const bf16x8*>(
+            &lds_b[buf][row * BK + swz(row, col)]);
+      }
+#pragma unroll
+      for (int mi = 0; mi < M_FRAGS; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < N_FRAGS; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              __builtin_bit_cast(short8v, afrag[mi]),
+              __builtin_bit_cast(short8v, bfrag[ni]), acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: bias + act, write NHWC ----
+  // C/D layout (16x16): col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+  for (int mi = 0; mi < M_FRAGS; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < N_FRAGS; ++ni) {
+      const int n = n0 + wn + ni * 16 + (lane & 15);
+      if (n >= K) continue;
+      const float bv = bias ? bias[n] : 0.f;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int m = m0 + wm + mi * 16 + (lane >> 4) * 4 + reg;
+        if (m >= M) continue;
+        const float val = act_fn<ACT>(acc[mi][ni][reg] + bv);
+        out[(long)m * K + n] = (bf16)val;
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// =====================================================================
+// Host launcher: x [B,C,H,W] channels_last, w [K,C,R,S] channels_last,
+// bias fp32 or undefined, act: 0 none / 1 ELU / 2 LeakyReLU(0.1).
+// =====================================================================
+at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
+                      long stride, long pad, long act) {
+  TORCH_CHECK(x.is_cuda() && w.is_cuda());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+              w.scalar_type() == at::kBFloat16, "bf16 only");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "x must be channels_last");
+  TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "w must be channels_last");
+  const int B = x.size(0), C = x.size(1), IH = x.size(2), IW = x.size(3);
+  const int K = w.size(0), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(w.size(1) == C);
+  TORCH_CHECK(C % 8 == 0, "conv2d_fwd needs C % 8 == 0, got ", C);
+  const int OH = (IH + 2 * (int)pad - R) / (int)stride + 1;
+  const int OW = (IW + 2 * (int)pad - S) / (int)stride + 1;
+  const long M = (long)B * OH * OW;
+
+  constexpr int BM = 128, BK = 64;
+  TORCH_CHECK((R * S * C) % BK == 0, "R*S*C must be a multiple of ", BK);
+
+  auto out = at::empty({B, K, OH, OW},
+                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  const float* bptr = nullptr;
+  at::Tensor bias_f;
+  if (bias.defined() && bias.numel()) {
+    bias_f = bias.to(at::kFloat).contiguous();
+    bptr = bias_f.data_ptr<float>();
+  }
+
+  const bf16* xp = reinterpret_cast<const bf16*>(x.data_ptr());
+  const bf16* wp = reinterpret_cast<const bf16*>(w.data_ptr());
+  bf16* op = reinterpret_cast<bf16*>(out.data_ptr());
+
+  const int BN = (K >= 128) ? 128 : 64;
+  const int n_tiles_n = (K + BN - 1) / BN;
+  const long n_blocks = ((M + BM - 1) / BM) * n_tiles_n;
+  const dim3 grid((unsigned)n_blocks), block(256);
+
+#define LAUNCH(BN_, ACT_)                                                  \
+  hipLaunchKernelGGL((conv_fwd_mfma_kernel<BM, BN_, BK, ACT_>), grid,      \
+                     block, 0, deepof_stream(), xp, wp, bptr, op, B, IH,   \
+                     IW, C, K, R, S, OH, OW, (int)stride, (int)pad,        \
+                     n_tiles_n)
+  if (BN == 128) {
+    if (act == 1) LAUNCH(128, 1);
+    else if (act == 2) LAUNCH(128, 2);
+    else LAUNCH(128, 0);
+  } else {
+    if (act == 1) LAUNCH(64, 1);
+    else if (act == 2) LAUNCH(64, 2);
+    else LAUNCH(64, 0);
+  }
+#undef LAUNCH
+  return out;
+}

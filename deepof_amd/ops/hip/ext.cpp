@@ -20,6 +20,8 @@ at::Tensor epe_sum(at::Tensor f, at::Tensor g);
 at::Tensor correlation_forward(at::Tensor f1, at::Tensor f2, long md);
 std::vector<at::Tensor> correlation_backward(at::Tensor gout, at::Tensor f1,
                                              at::Tensor f2, long md);
+at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
+                      long stride, long pad, long act);
 void fused_adam(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                 std::vector<at::Tensor> exp_avgs,
                 std::vector<at::Tensor> exp_avg_sqs, double lr, double beta1,
@@ -39,4 +41,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("correlation_forward", &correlation_forward, "cost volume fwd");
   m.def("correlation_backward", &correlation_backward, "cost volume bwd");
   m.def("fused_adam", &fused_adam, "multi-tensor Adam step");
+  m.def("conv2d_fwd", &conv2d_fwd,
+        "MFMA implicit-GEMM conv + bias + act (NHWC bf16)");
 }

@@ -23,6 +23,7 @@ ext = CUDAExtension(
         os.path.join(SRC, "loss_kernels.hip"),
         os.path.join(SRC, "correlation.hip"),
         os.path.join(SRC, "adam.hip"),
+        os.path.join(SRC, "conv_mfma.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
