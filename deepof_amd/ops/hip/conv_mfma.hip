@@ -56,17 +56,20 @@ __device__ inline int swz(int row, int col) {
 // Grid: (ceil(M/BM) * ceil(K/BN)) blocks, 256 threads.
 // Requires C % CBK == 0 where CBK = min(C, BK).
 // ---------------------------------------------------------------------
-template <int BM, int BN, int BK, int ACT>
+template <int BM, int BN, int BK, int ACT, bool GLDS>
 __global__ __launch_bounds__(256)
 void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
                           const bf16* __restrict__ w,
                           const float* __restrict__ bias,
                           bf16* __restrict__ out,
+                          const bf16* __restrict__ zero_page,
                           int B, int IH, int IW, int C,
                           int K, int R, int S, int OH, int OW,
                           int stride, int pad, int n_tiles_n) {
-  __shared__ bf16 lds_a[2][BM * BK];
-  __shared__ bf16 lds_b[2][BN * BK];
+  __shared__ bf16 lds_all[2 * (BM + BN) * BK];
+  // pointer-array init from addrspace(3) is rejected; index arithmetic
+#define LDS_A(buf) (lds_all + (buf) * BM * BK)
+#define LDS_B(buf) (lds_all + 2 * BM * BK + (buf) * BN * BK)
 
   const int tile_m = blockIdx.x / n_tiles_n;
   const int tile_n = blockIdx.x % n_tiles_n;
@@ -84,42 +87,101 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
   const int wm = (wid / WAVES_N) * (BM / WAVES_M);
   const int wn = (wid % WAVES_N) * 64;
 
-  // ---- per-thread A-load geometry (BM x BK tile, 16B per lane-load) ----
-  // thread t loads rows; each row of A needs BK*2 = 128 B = 8 lane-loads.
-  // 256 threads * 16 B = 4 KiB per pass; BM*BK*2 = 16 KiB -> 4 passes.
-  // Layout: loader covers (row, kchunk): row = t / (BK/8), kchunk = t % (BK/8)
-  constexpr int KCHUNKS = BK / 8;              // 16-B chunks per row
-  constexpr int ROWS_PER_PASS = 256 / KCHUNKS; // rows loaded per pass
-  const int lrow = tid / KCHUNKS;
-  const int lk = (tid % KCHUNKS) * 8;
-
-  // precompute this thread's A pixel coords for each pass
-  int a_b[BM / ROWS_PER_PASS], a_oy[BM / ROWS_PER_PASS],
-      a_ox[BM / ROWS_PER_PASS];
-#pragma unroll
-  for (int p = 0; p < BM / ROWS_PER_PASS; ++p) {
-    const int m = m0 + lrow + p * ROWS_PER_PASS;
-    const int mm = m < M ? m : M - 1;
-    a_ox[p] = mm % OW;
-    a_oy[p] = (mm / OW) % OH;
-    a_b[p] = mm / (OW * OH);
-    if (m >= M) a_b[p] = -1;  // padded row -> zeros
-  }
+  const int n_stages = (R * S * C) / BK; // total K stages
 
   f32x4 acc[M_FRAGS][N_FRAGS] = {};
 
-  const int CBK = (C < BK) ? C : BK;     // channels per k-stage chunk
-  const int taps_per_stage = BK / CBK;   // (r,s) taps packed per stage
-  const int n_stages = (R * S * C) / BK; // total K stages
-  const long ihw = (long)IH * IW;
+  // =====================================================================
+  // GLDS staging path (requires C % BK == 0): each wave issues 1-KiB
+  // global_load_lds_dwordx4 groups (8 rows x 128 B); the T2 swizzle is
+  // applied on the per-lane SOURCE address (rule 21 - LDS dest stays
+  // lane-linear), OOB taps and padded rows read a 128-B zero page.
+  // =====================================================================
+  // per-lane fixed geometry
+  const int g_row_in_grp = lane >> 3;          // 0..7
+  const int g_colb = (lane & 7) * 8;           // element col of this lane
+  constexpr int A_GROUPS = BM / 8;             // 1-KiB groups in A tile
+  constexpr int B_GROUPS = BN / 8;
 
-  // ---- staging helpers ----
-  auto stage = [&](int stage_idx, int buf) {
-    // reduction range [stage_idx*BK, stage_idx*BK + BK) of (r,s,c)
-    // A tile
+  // A pixel coords per group handled by this wave (row fixed per group)
+  int ga_iy[A_GROUPS / 4], ga_ix[A_GROUPS / 4];
+  long ga_base[A_GROUPS / 4];  // b*IH*IW offset or -1
+  if (GLDS) {
+#pragma unroll
+    for (int gi = 0; gi < A_GROUPS / 4; ++gi) {
+      const int g = wid + gi * 4;
+      const int row = g * 8 + g_row_in_grp;
+      const int m = m0 + row;
+      const int mm = m < M ? m : 0;
+      const int ox = mm % OW;
+      const int oy = (mm / OW) % OH;
+      const int bb = mm / (OW * OH);
+      ga_iy[gi] = oy * stride - pad;
+      ga_ix[gi] = ox * stride - pad;
+      ga_base[gi] = (m < M) ? (long)bb * IH * IW : -1;
+    }
+  }
+
+  auto stage_glds = [&](int stage_idx, int buf) {
+    const int rsc0 = stage_idx * BK;
+    const int rs = rsc0 / C;       // uniform: C % BK == 0
+    const int r = rs / S, ss = rs % S;
+    const int c0 = rsc0 % C;
+    const int col = g_colb ^ ((g_row_in_grp & 7) << 3);  // source swizzle
+#pragma unroll
+    for (int gi = 0; gi < A_GROUPS / 4; ++gi) {
+      const int g = wid + gi * 4;
+      const int iy = ga_iy[gi] + r;
+      const int ix = ga_ix[gi] + ss;
+      const bf16* src = zero_page;
+      if (ga_base[gi] >= 0 && iy >= 0 && iy < IH && ix >= 0 && ix < IW)
+        src = x + (ga_base[gi] + (long)iy * IW + ix) * C + c0 + col;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)
+              &LDS_A(buf)[g * 8 * BK],
+          16, 0, 0);
+    }
+#pragma unroll
+    for (int gi = 0; gi < B_GROUPS / 4; ++gi) {
+      const int g = wid + gi * 4;
+      const int n = n0 + g * 8 + g_row_in_grp;
+      const bf16* src = (n < K)
+          ? w + (long)n * (R * S * C) + rsc0 + col
+          : zero_page;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)
+              &LDS_B(buf)[g * 8 * BK],
+          16, 0, 0);
+    }
+  };
+
+  // =====================================================================
+  // generic staging path (any C % 8 == 0): plain loads + ds_write
+  // =====================================================================
+  constexpr int KCHUNKS = BK / 8;
+  constexpr int ROWS_PER_PASS = 256 / KCHUNKS;
+  const int lrow = tid / KCHUNKS;
+  const int lk = (tid % KCHUNKS) * 8;
+  int a_b[BM / ROWS_PER_PASS], a_oy[BM / ROWS_PER_PASS],
+      a_ox[BM / ROWS_PER_PASS];
+  if (!GLDS) {
 #pragma unroll
     for (int p = 0; p < BM / ROWS_PER_PASS; ++p) {
-      const int rsc = stage_idx * BK + lk;  // global reduction index
+      const int m = m0 + lrow + p * ROWS_PER_PASS;
+      const int mm = m < M ? m : M - 1;
+      a_ox[p] = mm % OW;
+      a_oy[p] = (mm / OW) % OH;
+      a_b[p] = mm / (OW * OH);
+      if (m >= M) a_b[p] = -1;
+    }
+  }
+
+  auto stage = [&](int stage_idx, int buf) {
+#pragma unroll
+    for (int p = 0; p < BM / ROWS_PER_PASS; ++p) {
+      const int rsc = stage_idx * BK + lk;
       const int c = rsc % C;
       const int rs = rsc / C;
       const int s = rs % S, r = rs / S;
@@ -133,9 +195,8 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
         }
       }
       const int row = lrow + p * ROWS_PER_PASS;
-      *reinterpret_cast<bf16x8*>(&lds_a[buf][row * BK + swz(row, lk)]) = v;
+      *reinterpret_cast<bf16x8*>(&LDS_A(buf)[row * BK + swz(row, lk)]) = v;
     }
-    // B tile: rows = output channel n, cols = reduction
 #pragma unroll
     for (int p = 0; p < BN / ROWS_PER_PASS; ++p) {
       const int n = n0 + lrow + p * ROWS_PER_PASS;
@@ -146,16 +207,18 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
         v = *reinterpret_cast<const bf16x8*>(src);
       }
       const int row = lrow + p * ROWS_PER_PASS;
-      *reinterpret_cast<bf16x8*>(&lds_b[buf][row * BK + swz(row, lk)]) = v;
+      *reinterpret_cast<bf16x8*>(&LDS_B(buf)[row * BK + swz(row, lk)]) = v;
     }
   };
 
-  stage(0, 0);
+  if (GLDS) stage_glds(0, 0); else stage(0, 0);
   __syncthreads();
 
   for (int st = 0; st < n_stages; ++st) {
     const int buf = st & 1;
-    if (st + 1 < n_stages) stage(st + 1, buf ^ 1);
+    if (st + 1 < n_stages) {
+      if (GLDS) stage_glds(st + 1, buf ^ 1); else stage(st + 1, buf ^ 1);
+    }
 
     // MFMA over this stage: BK reduction = BK/32 mfma k-steps
 #pragma unroll
@@ -168,14 +231,14 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
         const int row = wm + mi * 16 + (lane & 15);
         const int col = kk + (lane >> 4) * 8;
         afrag[mi] = *reinterpret_cast<const bf16x8*>(
-            &lds_a[buf][row * BK + swz(row, col)]);
+            &LDS_A(buf)[row * BK + swz(row, col)]);
       }
 #pragma unroll
       for (int ni = 0; ni < N_FRAGS; ++ni) {
         const int row = wn + ni * 16 + (lane & 15);
         const int col = kk + (lane >> 4) * 8;
         bfrag[ni] = *reinterpret_cast<const bf16x8*>(
-            &lds_b[buf][row * BK + swz(row, col)]);
+            &LDS_B(buf)[row * BK + swz(row, col)]);
       }
 #pragma unroll
       for (int mi = 0; mi < M_FRAGS; ++mi)
@@ -206,6 +269,8 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
       }
     }
   }
+#undef LDS_A
+#undef LDS_B
 }
 
 }  // namespace
@@ -252,20 +317,30 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
   const long n_blocks = ((M + BM - 1) / BM) * n_tiles_n;
   const dim3 grid((unsigned)n_blocks), block(256);
 
-#define LAUNCH(BN_, ACT_)                                                  \
-  hipLaunchKernelGGL((conv_fwd_mfma_kernel<BM, BN_, BK, ACT_>), grid,      \
-                     block, 0, deepof_stream(), xp, wp, bptr, op, B, IH,   \
-                     IW, C, K, R, S, OH, OW, (int)stride, (int)pad,        \
-                     n_tiles_n)
+  // 128-B zero page for OOB taps (glds cannot mask)
+  static at::Tensor zero_page;
+  if (!zero_page.defined() || zero_page.device() != x.device())
+    zero_page = at::zeros({64}, x.options());
+  const bf16* zp = reinterpret_cast<const bf16*>(zero_page.data_ptr());
+  const bool glds = (C % BK == 0);
+
+#define LAUNCH(BN_, ACT_, GLDS_)                                           \
+  hipLaunchKernelGGL((conv_fwd_mfma_kernel<BM, BN_, BK, ACT_, GLDS_>),     \
+                     grid, block, 0, deepof_stream(), xp, wp, bptr, op,    \
+                     zp, B, IH, IW, C, K, R, S, OH, OW, (int)stride,       \
+                     (int)pad, n_tiles_n)
+#define LAUNCH_ACT(BN_, GLDS_)                                             \
+  do {                                                                     \
+    if (act == 1) LAUNCH(BN_, 1, GLDS_);                                   \
+    else if (act == 2) LAUNCH(BN_, 2, GLDS_);                              \
+    else LAUNCH(BN_, 0, GLDS_);                                            \
+  } while (0)
   if (BN == 128) {
-    if (act == 1) LAUNCH(128, 1);
-    else if (act == 2) LAUNCH(128, 2);
-    else LAUNCH(128, 0);
+    if (glds) LAUNCH_ACT(128, true); else LAUNCH_ACT(128, false);
   } else {
-    if (act == 1) LAUNCH(64, 1);
-    else if (act == 2) LAUNCH(64, 2);
-    else LAUNCH(64, 0);
+    if (glds) LAUNCH_ACT(64, true); else LAUNCH_ACT(64, false);
   }
+#undef LAUNCH_ACT
 #undef LAUNCH
   return out;
 }
