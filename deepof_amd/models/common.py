@@ -29,12 +29,9 @@ def make_act(name: str) -> nn.Module:
 
 def conv(cin: int, cout: int, k: int = 3, stride: int = 1,
          act: str | None = "elu") -> nn.Module:
-    layers: list[nn.Module] = [
-        nn.Conv2d(cin, cout, k, stride=stride, padding=k // 2, bias=True)
-    ]
-    if act is not None:
-        layers.append(make_act(act))
-    return nn.Sequential(*layers) if len(layers) > 1 else layers[0]
+    from ..ops.conv import FusedConvAct
+
+    return FusedConvAct(cin, cout, k, stride, act)
 
 
 def deconv(cin: int, cout: int, act: str | None = "elu") -> nn.Module:
@@ -127,8 +124,12 @@ class FlowDecoder(nn.Module):
             if i < self.num_scales - 1:
                 self.upconvs.append(up(concat_ch, up_channels[i], up_factors[i], act))
                 uf = up(flow_channels, flow_channels, up_factors[i], None)
-                m = uf if isinstance(uf, (nn.ConvTranspose2d, nn.Conv2d)) else uf[0]
-                m._bilinear_init = True
+                if isinstance(uf, nn.ConvTranspose2d):
+                    uf._bilinear_init = True
+                elif isinstance(uf, nn.Sequential):
+                    uf[0]._bilinear_init = True
+                else:  # FusedConvAct (stride-1 refinement stage)
+                    uf.conv._bilinear_init = True
                 self.upflows.append(uf)
                 concat_ch = feat_channels[i + 1] + up_channels[i] + flow_channels
 

@@ -34,11 +34,12 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(8))) short short8v;
 
-// activation: 0 = none, 1 = ELU, 2 = LeakyReLU(0.1)
+// activation: 0 = none, 1 = ELU, 2 = LeakyReLU(0.1), 3 = ReLU
 template <int ACT>
 __device__ inline float act_fn(float x) {
   if (ACT == 1) return x > 0.f ? x : expf(x) - 1.f;
   if (ACT == 2) return x > 0.f ? x : 0.1f * x;
+  if (ACT == 3) return x > 0.f ? x : 0.f;
   return x;
 }
 
@@ -333,6 +334,7 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
   do {                                                                     \
     if (act == 1) LAUNCH(BN_, 1, GLDS_);                                   \
     else if (act == 2) LAUNCH(BN_, 2, GLDS_);                              \
+    else if (act == 3) LAUNCH(BN_, 3, GLDS_);                              \
     else LAUNCH(BN_, 0, GLDS_);                                            \
   } while (0)
   if (BN == 128) {
