@@ -22,6 +22,13 @@ class FusedAdam(Optimizer):
         defaults = dict(lr=lr, betas=betas, eps=eps,
                         weight_decay=weight_decay)
         super().__init__(params, defaults)
+        # cached device chunk table + pinned hyper buffer (hipGraph-safe:
+        # the captured step re-reads [lr, bias1, bias2] from pinned host
+        # memory at every replay)
+        self._table = None
+        self._table_sig = None
+        self._hyper_pin = None
+        self._hyper_dev = None
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -56,10 +63,23 @@ class FusedAdam(Optimizer):
             if params[0].is_cuda:
                 from ..ops.functional import require_hip
 
-                require_hip().fused_adam(
-                    params, grads, exp_avgs, exp_avg_sqs,
-                    group["lr"], beta1, beta2, group["eps"],
-                    group["weight_decay"], bias1, bias2,
+                hip = require_hip()
+                sig = (len(params), params[0].data_ptr(),
+                       grads[0].data_ptr(), params[-1].data_ptr())
+                if self._table is None or self._table_sig != sig:
+                    self._table = hip.build_adam_table(
+                        params, grads, exp_avgs, exp_avg_sqs)
+                    self._table_sig = sig
+                    self._n_chunks = self._table.numel() // 40  # sizeof(ChunkInfo)
+                    self._hyper_pin = torch.zeros(3, pin_memory=True)
+                    self._hyper_dev = torch.zeros(3, device=params[0].device)
+                self._hyper_pin[0] = group["lr"]
+                self._hyper_pin[1] = bias1
+                self._hyper_pin[2] = bias2
+                self._hyper_dev.copy_(self._hyper_pin, non_blocking=True)
+                hip.fused_adam_table(
+                    self._table, self._n_chunks, self._hyper_dev,
+                    beta1, beta2, group["eps"], group["weight_decay"],
                 )
             else:
                 if group["weight_decay"] != 0.0:  # L2 (pre-moment, like TF)

@@ -155,10 +155,15 @@ class _FusedUnsupLoss(torch.autograd.Function):
         flow_raw, img1, img2 = ctx.saved_tensors
         flow_scale, eps, alpha_c, alpha_s = ctx.params
         hip = require_hip()
+        # upstream grads stay on device (no .item() sync; hipGraph-safe)
+        dev = flow_raw.device
+        def _g(t):
+            return (t.float().reshape(1) if isinstance(t, torch.Tensor)
+                    else torch.zeros(1, device=dev))
+        gs = torch.cat([_g(g_photo), _g(g_u), _g(g_v)])
         gflow = hip.unsup_loss_backward(
             flow_raw.contiguous(), img1.contiguous(), img2.contiguous(),
-            flow_scale, eps, alpha_c, alpha_s,
-            float(g_photo), float(g_u), float(g_v),
+            flow_scale, eps, alpha_c, alpha_s, gs,
         )
         return gflow, None, None, None, None, None, None, None
 

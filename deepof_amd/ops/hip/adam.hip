@@ -28,10 +28,14 @@ struct ChunkInfo {
   int n;
 };
 
+// hyper: optional device [lr, bias1, bias2] so a captured hipGraph can
+// replay with per-step bias correction (host updates the pinned source).
 __global__ void fused_adam_kernel(const ChunkInfo* __restrict__ chunks,
                                   int n_chunks, float lr, float beta1,
                                   float beta2, float eps, float wd,
-                                  float bias1, float bias2) {
+                                  float bias1, float bias2,
+                                  const float* __restrict__ hyper) {
+  if (hyper) { lr = hyper[0]; bias1 = hyper[1]; bias2 = hyper[2]; }
   const float step_size = lr / bias1;
   const float inv_sqrt_bias2 = rsqrtf(bias2);
   for (int ci = blockIdx.y; ci < n_chunks; ci += gridDim.y) {
@@ -78,6 +82,46 @@ __global__ void fused_adam_kernel(const ChunkInfo* __restrict__ chunks,
 
 }  // namespace
 
+// Build the device-side chunk table once; pointers stay valid across
+// steps (params/moments update in place), so the step itself is a pure
+// device launch — hipGraph-capturable.
+at::Tensor build_adam_table(std::vector<at::Tensor> params,
+                            std::vector<at::Tensor> grads,
+                            std::vector<at::Tensor> exp_avgs,
+                            std::vector<at::Tensor> exp_avg_sqs) {
+  std::vector<ChunkInfo> chunks;
+  for (size_t t = 0; t < params.size(); ++t) {
+    TORCH_CHECK(params[t].scalar_type() == at::kFloat);
+    auto* p = params[t].data_ptr<float>();
+    auto* g = grads[t].data_ptr<float>();
+    auto* m = exp_avgs[t].data_ptr<float>();
+    auto* v = exp_avg_sqs[t].data_ptr<float>();
+    const long n = params[t].numel();
+    for (long off = 0; off < n; off += CHUNK) {
+      const int len = (int)std::min((long)CHUNK, n - off);
+      chunks.push_back({p + off, g + off, m + off, v + off, len});
+    }
+  }
+  auto cpu = at::from_blob(chunks.data(),
+                           {(long)(chunks.size() * sizeof(ChunkInfo))},
+                           at::TensorOptions().dtype(at::kByte)).clone();
+  return cpu.to(params[0].device());
+}
+
+void fused_adam_table(at::Tensor table, long n_chunks, at::Tensor hyper,
+                      double beta1, double beta2, double eps, double wd) {
+  TORCH_CHECK(hyper.is_cuda() && hyper.scalar_type() == at::kFloat &&
+              hyper.numel() >= 3);
+  const int blocks_x = (CHUNK / 4 + BLOCK - 1) / BLOCK;
+  const dim3 grid(std::min(blocks_x, 2048), std::min((int)n_chunks, 64));
+  hipLaunchKernelGGL(fused_adam_kernel, grid, dim3(BLOCK), 0,
+                     deepof_stream(),
+                     reinterpret_cast<const ChunkInfo*>(table.data_ptr()),
+                     (int)n_chunks, 0.f, (float)beta1, (float)beta2,
+                     (float)eps, (float)wd, 1.f, 1.f,
+                     hyper.data_ptr<float>());
+}
+
 void fused_adam(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                 std::vector<at::Tensor> exp_avgs,
                 std::vector<at::Tensor> exp_avg_sqs, double lr, double beta1,
@@ -118,5 +162,5 @@ void fused_adam(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                      reinterpret_cast<const ChunkInfo*>(table.data_ptr()),
                      (int)chunks.size(), (float)lr, (float)beta1,
                      (float)beta2, (float)eps, (float)wd, (float)bias1,
-                     (float)bias2);
+                     (float)bias2, (const float*)nullptr);
 }

@@ -11,8 +11,7 @@ std::vector<at::Tensor> unsup_loss_forward(at::Tensor flow, at::Tensor img1,
                                            double alpha_s, bool want_recon);
 at::Tensor unsup_loss_backward(at::Tensor flow, at::Tensor img1,
                                at::Tensor img2, double scale, double eps,
-                               double alpha_c, double alpha_s, double g_photo,
-                               double g_u, double g_v);
+                               double alpha_c, double alpha_s, at::Tensor gs);
 at::Tensor resize_bilinear(at::Tensor x, long oh, long ow);
 at::Tensor lrn_forward(at::Tensor x, long radius, double bias, double alpha,
                        double beta);
@@ -22,6 +21,12 @@ std::vector<at::Tensor> correlation_backward(at::Tensor gout, at::Tensor f1,
                                              at::Tensor f2, long md);
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                       long stride, long pad, long act);
+at::Tensor build_adam_table(std::vector<at::Tensor> params,
+                            std::vector<at::Tensor> grads,
+                            std::vector<at::Tensor> exp_avgs,
+                            std::vector<at::Tensor> exp_avg_sqs);
+void fused_adam_table(at::Tensor table, long n_chunks, at::Tensor hyper,
+                      double beta1, double beta2, double eps, double wd);
 void fused_adam(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                 std::vector<at::Tensor> exp_avgs,
                 std::vector<at::Tensor> exp_avg_sqs, double lr, double beta1,
@@ -41,6 +46,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("correlation_forward", &correlation_forward, "cost volume fwd");
   m.def("correlation_backward", &correlation_backward, "cost volume bwd");
   m.def("fused_adam", &fused_adam, "multi-tensor Adam step");
+  m.def("build_adam_table", &build_adam_table, "pack Adam chunk table");
+  m.def("fused_adam_table", &fused_adam_table, "Adam step from table");
   m.def("conv2d_fwd", &conv2d_fwd,
         "MFMA implicit-GEMM conv + bias + act (NHWC bf16)");
 }

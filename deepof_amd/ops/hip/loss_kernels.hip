@@ -215,9 +215,10 @@ __global__ void unsup_loss_bwd_kernel(
     const float* __restrict__ flow,
     const T* __restrict__ img1, const T* __restrict__ img2,
     float* __restrict__ gflow,
+    const float* __restrict__ gs,  // device [g_photo, g_u, g_v]
     int B, int C, int H, int W, float scale,
-    float eps2, float alpha_c, float alpha_s, int bw,
-    float g_photo, float g_u, float g_v) {
+    float eps2, float alpha_c, float alpha_s, int bw) {
+  const float g_photo = gs[0], g_u = gs[1], g_v = gs[2];
   const int idx = blockIdx.x * blockDim.x + threadIdx.x;
   const int hw = H * W;
   if (idx >= B * hw) return;
@@ -441,7 +442,9 @@ std::vector<at::Tensor> unsup_loss_forward(at::Tensor flow, at::Tensor img1,
 at::Tensor unsup_loss_backward(at::Tensor flow, at::Tensor img1,
                                at::Tensor img2, double scale, double eps,
                                double alpha_c, double alpha_s,
-                               double g_photo, double g_u, double g_v) {
+                               at::Tensor gs) {
+  TORCH_CHECK(gs.is_cuda() && gs.scalar_type() == at::kFloat &&
+              gs.numel() == 3);
   const int B = img1.size(0), C = img1.size(1), H = img1.size(2), W = img1.size(3);
   const int bw_full = (int)ceil(H * 0.1);
   const int bw = (H - 2 * bw_full > 0 && W - 2 * bw_full > 0) ? bw_full : 0;
@@ -455,9 +458,9 @@ at::Tensor unsup_loss_backward(at::Tensor flow, at::Tensor img1,
                        deepof_stream(),
                        flow.data_ptr<float>(), img1.data_ptr<scalar_t>(),
                        img2.data_ptr<scalar_t>(), gflow.data_ptr<float>(),
+                       gs.data_ptr<float>(),
                        B, C, H, W, (float)scale, eps2, (float)alpha_c,
-                       (float)alpha_s, bw, (float)g_photo, (float)g_u,
-                       (float)g_v);
+                       (float)alpha_s, bw);
   });
   return gflow;
 }
