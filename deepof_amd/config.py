@@ -35,6 +35,9 @@ class Config:
     model: str = "flownets"              # flownets|flownetc|vgg16|inception_v3
     activation: str = "elu"
 
+    augment: bool = False                # GPU geo+photo augmentation
+    vgg_init: str = ""                   # path to vgg16_weights.npz warm start
+
     # loss (SURVEY §2.5 defaults: chairs FlowNetS v0)
     epsilon: float = 1e-4
     alpha_c: float = 0.25
@@ -64,6 +67,8 @@ class Config:
     log_interval: int = 50
     eval_interval_epochs: int = 1
     resume: bool = True
+
+    nan_restart_limit: int = 3           # auto-restarts from ckpt on NaN
 
     # action head (UCF101 joint training)
     action_classes: int = 0              # >0 enables the action head

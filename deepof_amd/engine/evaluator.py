@@ -43,7 +43,11 @@ def predict_flow(model, img1_raw, img2_raw, mean_bgr, flow_scale_finest,
 
 @torch.no_grad()
 def evaluate_aee(model, loader, mean_bgr, flow_scale_finest, device,
-                 dataset: str = "flying_chairs", max_batches=None) -> float:
+                 dataset: str = "flying_chairs", max_batches=None,
+                 dump_dir: str | None = None, dump_every: int = 10) -> float:
+    """AEE over a loader; optionally dump flow color maps, warped
+    frames and predicted .flo files (parity with the reference's eval
+    artifacts, flyingChairsTrain.py:272-291)."""
     model.eval()
     total = 0.0
     count = 0
@@ -57,5 +61,31 @@ def evaluate_aee(model, loader, mean_bgr, flow_scale_finest, device,
                             dataset, gt_size=tuple(gt.shape[-2:]))
         total += float(ops.endpoint_error_sum(pred, gt))
         count += gt.shape[0] * gt.shape[-2] * gt.shape[-1]
+        if dump_dir is not None and i % dump_every == 0:
+            _dump_artifacts(dump_dir, i, img1, img2, pred, gt)
     model.train()
     return total / max(count, 1)
+
+
+def _dump_artifacts(dump_dir, batch_idx, img1, img2, pred, gt):
+    import os
+
+    import numpy as np
+    from PIL import Image
+
+    from ..utils import flow_to_color, write_flo
+
+    os.makedirs(dump_dir, exist_ok=True)
+    p = pred[0].permute(1, 2, 0).cpu().numpy()
+    g = gt[0].permute(1, 2, 0).cpu().numpy()
+    write_flo(os.path.join(dump_dir, f"b{batch_idx:04d}_pred.flo"), p)
+    Image.fromarray(flow_to_color(p)).save(
+        os.path.join(dump_dir, f"b{batch_idx:04d}_pred.jpg"))
+    Image.fromarray(flow_to_color(g)).save(
+        os.path.join(dump_dir, f"b{batch_idx:04d}_gt.jpg"))
+    # warped frame 2 (reconstruction of frame 1) at image resolution
+    flow_img = ops.resize_bilinear(pred[:1], img1.shape[-2], img1.shape[-1])
+    recon = ops.warp_bilinear(img2[:1].float(), flow_img)
+    rec = recon[0].permute(1, 2, 0).cpu().numpy()[:, :, ::-1]  # BGR->RGB
+    Image.fromarray(np.clip(rec, 0, 255).astype(np.uint8)).save(
+        os.path.join(dump_dir, f"b{batch_idx:04d}_warped.jpg"))

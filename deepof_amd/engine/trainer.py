@@ -62,8 +62,20 @@ class Trainer:
         model_kwargs = {"act": cfg.activation}
         if cfg.model == "inception_v3" and cfg.time_step > 2:
             model_kwargs["time_step"] = cfg.time_step
+        if cfg.model in ("st_single", "st_baseline"):
+            model_kwargs["input_hw"] = tuple(cfg.crop_size or cfg.image_size)
+            model_kwargs["num_classes"] = cfg.action_classes or 101
         self.model, self.flow_scales, default_w = build_model(
             cfg.model, **model_kwargs)
+        if cfg.vgg_init:
+            from ..utils.warmstart import load_vgg16_npz
+
+            enc = getattr(self.model, "encoder", None) or getattr(
+                self.model, "spatial", None)
+            n = load_vgg16_npz(enc, cfg.vgg_init)
+            if self.rank == 0:
+                print(f"[deepof] VGG16 warm start: {n} tensors from "
+                      f"{cfg.vgg_init}")
         self.loss_weights = cfg.loss_weights or default_w
         self.model.to(self.device)
         if cfg.channels_last:
@@ -158,27 +170,66 @@ class Trainer:
     # -- the step ---------------------------------------------------------
     def train_step(self, batch) -> dict:
         cfg = self.cfg
-        img1 = batch["img1"].to(self.device, non_blocking=True)
-        img2 = batch["img2"].to(self.device, non_blocking=True)
         from ..losses.unsup import preprocess_images
 
-        x1 = preprocess_images(img1.float(), self.mean_bgr)
-        x2 = preprocess_images(img2.float(), self.mean_bgr)
-        x = torch.cat([x1, x2], dim=1)
-        if cfg.channels_last:
-            x = x.to(memory_format=torch.channels_last)
+        use_bf16 = cfg.precision == "bf16" and self.device.type == "cuda"
+        parts = {}
 
-        use_bf16 = cfg.precision == "bf16" and x.is_cuda
-        with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16):
-            flows = self.model(x)
-        res = self.unsup_loss(flows, img1, img2)
-        total = cfg.photo_weight * res["total"]
-        parts = {"unsup": float(res["total"].detach())}
-        if self.guided_loss is not None and "flow" in batch:
-            gt = batch["flow"].to(self.device, non_blocking=True)
-            g = self.guided_loss(flows, gt)
-            total = total + cfg.guided_weight * g["total"]
-            parts["guided"] = float(g["total"].detach())
+        if "volume" in batch:  # Sintel multi-frame mode
+            vol = batch["volume"].to(self.device, non_blocking=True)
+            from ..losses import MultiFrameUnsupLoss
+
+            if not hasattr(self, "_mf_loss"):
+                self._mf_loss = MultiFrameUnsupLoss(
+                    self.flow_scales, self.loss_weights, self.mean_bgr,
+                    cfg.epsilon, cfg.alpha_c, cfg.alpha_s, cfg.lambda_smooth)
+            mean = torch.as_tensor(self.mean_bgr, dtype=torch.float32,
+                                   device=vol.device)
+            T = vol.shape[1] // 3
+            x = (vol.float() - mean.repeat(T).view(1, -1, 1, 1)) / 255.0
+            with torch.autocast("cuda", dtype=torch.bfloat16,
+                                enabled=use_bf16):
+                flows = self.model(x)
+            res = self._mf_loss(flows, vol)
+            total = res["total"]
+            parts["unsup"] = float(total.detach())
+        else:
+            img1 = batch["img1"].to(self.device, non_blocking=True)
+            img2 = batch["img2"].to(self.device, non_blocking=True)
+            geo1, geo2 = img1.float(), img2.float()
+            net1, net2 = geo1, geo2
+            if cfg.augment:
+                from ..utils.augment import augment_pair
+
+                geo1, geo2, net1, net2 = augment_pair(geo1, geo2)
+            x1 = preprocess_images(net1, self.mean_bgr)
+            x2 = preprocess_images(net2, self.mean_bgr)
+            x = torch.cat([x1, x2], dim=1)
+            if cfg.channels_last:
+                x = x.to(memory_format=torch.channels_last)
+
+            with torch.autocast("cuda", dtype=torch.bfloat16,
+                                enabled=use_bf16):
+                out = self.model(x)
+            logits = None
+            flows = out
+            if isinstance(out, tuple):  # joint flow+action models
+                flows, logits = out
+            res = self.unsup_loss(flows, geo1, geo2)
+            total = cfg.photo_weight * res["total"]
+            parts["unsup"] = float(res["total"].detach())
+            if self.guided_loss is not None and "flow" in batch:
+                gt = batch["flow"].to(self.device, non_blocking=True)
+                g = self.guided_loss(flows, gt)
+                total = total + cfg.guided_weight * g["total"]
+                parts["guided"] = float(g["total"].detach())
+            if logits is not None and "label" in batch:
+                labels = batch["label"].to(self.device, non_blocking=True)
+                ce = torch.nn.functional.cross_entropy(logits.float(), labels)
+                total = total + cfg.action_weight * ce
+                parts["action_ce"] = float(ce.detach())
+                parts["action_acc"] = float(
+                    (logits.argmax(1) == labels).float().mean())
 
         total.backward()
         if isinstance(self.model, BucketedDataParallel):
@@ -215,13 +266,27 @@ class Trainer:
         )
         max_epochs = max_epochs or cfg.max_epochs
         steps_done = 0
+        nan_restarts = 0
         while self.epoch < max_epochs:
             if hasattr(loader, "sampler") and hasattr(loader.sampler, "set_epoch"):
                 loader.sampler.set_epoch(self.epoch)
             t0 = time.time()
             n_imgs = 0
             for i, batch in enumerate(loader):
-                parts = self.train_step(batch)
+                try:
+                    parts = self.train_step(batch)
+                except FloatingPointError as e:
+                    # divergence guard (reference asserts and dies,
+                    # flyingChairsTrain.py:203); here: restart from the
+                    # latest checkpoint, bounded retries
+                    nan_restarts += 1
+                    if (nan_restarts > cfg.nan_restart_limit
+                            or not self.try_resume()):
+                        raise
+                    self._zero_grads()
+                    print(f"[deepof] {e}; restarted from checkpoint "
+                          f"({nan_restarts}/{cfg.nan_restart_limit})")
+                    break
                 n_imgs += batch["img1"].shape[0] * self.world
                 steps_done += 1
                 if self.global_step % cfg.log_interval == 0:

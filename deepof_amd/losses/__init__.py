@@ -1,4 +1,6 @@
-from .unsup import MultiScaleUnsupLoss, preprocess_images
+from .unsup import (MultiFrameUnsupLoss, MultiScaleUnsupLoss,
+                    preprocess_images)
 from .guided import MultiScaleGuidedLoss
 
-__all__ = ["MultiScaleUnsupLoss", "MultiScaleGuidedLoss", "preprocess_images"]
+__all__ = ["MultiScaleUnsupLoss", "MultiFrameUnsupLoss",
+           "MultiScaleGuidedLoss", "preprocess_images"]

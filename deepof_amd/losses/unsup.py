@@ -100,3 +100,64 @@ class MultiScaleUnsupLoss(nn.Module):
         if recon is not None:
             out["recon"] = recon
         return out
+
+
+class MultiFrameUnsupLoss(nn.Module):
+    """Multi-frame (Sintel volume) unsupervised loss.
+
+    The reference's loss_interp_multi warps frame t+1 onto frame t for
+    every consecutive pair inside one graph op
+    (/root/reference/sintelWrapFlow.py:492-630, channel->flow index map
+    :543, source channel c+3 :565).  Here each pair reuses the fused
+    per-scale HIP loss on channel slices: flow pair t = channels
+    [2t, 2t+1], images frame t = channels [3t : 3t+3].  Pair losses are
+    averaged.  LRN is applied per frame (the reference LRN's
+    depth_radius=4 window leaks across frame boundaries in the stacked
+    volume — a quirk not replicated).
+    """
+
+    def __init__(self, flow_scales, loss_weights,
+                 mean_bgr=DATASET_MEANS["sintel"], epsilon=1e-4,
+                 alpha_c=0.3, alpha_s=0.3, lambda_smooth=0.0):
+        super().__init__()
+        self.flow_scales = flow_scales
+        self.loss_weights = loss_weights
+        self.mean_bgr = mean_bgr
+        self.epsilon = epsilon
+        self.alpha_c = alpha_c
+        self.alpha_s = alpha_s
+        self.lambda_smooth = lambda_smooth
+
+    def forward(self, flows: list, volume_raw: torch.Tensor):
+        """flows: raw [B, 2(T-1), h_k, w_k] finest first; volume_raw:
+        [B, 3T, H, W] 0-255 BGR."""
+        b, ct, H, W = volume_raw.shape
+        T = ct // 3
+        assert flows[0].shape[1] == 2 * (T - 1)
+        mean = torch.as_tensor(self.mean_bgr, dtype=torch.float32,
+                               device=volume_raw.device)
+        vol = (volume_raw.float() - mean.repeat(T).view(1, ct, 1, 1)) / 255.0
+        with torch.no_grad():
+            frames = [ops.lrn(vol[:, 3 * t: 3 * t + 3].contiguous())
+                      for t in range(T)]
+
+        total = None
+        scale_losses = []
+        for k, flow in enumerate(flows):
+            h, w = flow.shape[-2:]
+            with torch.no_grad():
+                pyr = [ops.resize_bilinear(f, h, w) for f in frames]
+            pair_total = None
+            for t in range(T - 1):
+                res = ops.unsup_loss_scale(
+                    flow[:, 2 * t: 2 * t + 2].float(), pyr[t], pyr[t + 1],
+                    self.flow_scales[k], self.epsilon, self.alpha_c,
+                    self.alpha_s, self.lambda_smooth,
+                )
+                pair_total = res["total"] if pair_total is None \
+                    else pair_total + res["total"]
+            loss_k = pair_total / (T - 1)
+            scale_losses.append(loss_k)
+            term = self.loss_weights[k] * loss_k
+            total = term if total is None else total + term
+        return {"total": total, "scales": scale_losses}

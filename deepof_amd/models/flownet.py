@@ -46,20 +46,23 @@ class FlowNetS(nn.Module):
         )
         init_flow_module(self)
 
-    def forward(self, x: torch.Tensor) -> list[torch.Tensor]:
-        """x: [B, 6, H, W] (concat of the normalized image pair).
-
-        Returns raw flow predictions FINEST FIRST: [pr1 ... pr6]
-        (pr1 at H/2 x W/2), to be scaled by FLOW_SCALES[k].
-        """
+    def encode(self, x: torch.Tensor) -> list[torch.Tensor]:
+        """Contracting stack; returns [c6, c5, c4, c3, c2, c1]."""
         c1 = self.conv1(x)
         c2 = self.conv2(c1)
         c3 = self.conv3_2(self.conv3_1(c2))
         c4 = self.conv4_2(self.conv4_1(c3))
         c5 = self.conv5_2(self.conv5_1(c4))
         c6 = self.conv6_2(self.conv6_1(c5))
-        flows_coarse_first = self.decoder([c6, c5, c4, c3, c2, c1])
-        return flows_coarse_first[::-1]
+        return [c6, c5, c4, c3, c2, c1]
+
+    def forward(self, x: torch.Tensor) -> list[torch.Tensor]:
+        """x: [B, 6, H, W] (concat of the normalized image pair).
+
+        Returns raw flow predictions FINEST FIRST: [pr1 ... pr6]
+        (pr1 at H/2 x W/2), to be scaled by FLOW_SCALES[k].
+        """
+        return self.decoder(self.encode(x))[::-1]
 
 
 class FlowNetC(nn.Module):

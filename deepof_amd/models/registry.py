@@ -9,6 +9,7 @@ from __future__ import annotations
 
 from .flownet import FLOW_SCALES, FlowNetC, FlowNetS
 from .inception import INCEPTION_FLOW_SCALES, InceptionFlow
+from .ucf101 import STBaseline, STSingle
 from .vgg16 import VGG_FLOW_SCALES, VGG16Flow
 
 MODEL_REGISTRY = {
@@ -26,6 +27,16 @@ MODEL_REGISTRY = {
         "ctor": VGG16Flow,
         "flow_scales": VGG_FLOW_SCALES,
         "loss_weights": [7.0, 5.0, 3.0, 3.0, 1.0],
+    },
+    "st_single": {
+        "ctor": STSingle,
+        "flow_scales": VGG_FLOW_SCALES,
+        "loss_weights": [16.0, 8.0, 4.0, 2.0, 1.0],
+    },
+    "st_baseline": {
+        "ctor": STBaseline,
+        "flow_scales": FLOW_SCALES,
+        "loss_weights": [16.0, 8.0, 4.0, 2.0, 1.0, 1.0],
     },
     "inception_v3": {
         "ctor": InceptionFlow,

