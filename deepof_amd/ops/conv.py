@@ -56,12 +56,31 @@ class _FusedConvFn(torch.autograd.Function):
                                   torch.full_like(y, 0.1))
         elif act_code == 3:
             gy = gy * (y > 0).to(gy.dtype)
-        gx, gw, gb = torch.ops.aten.convolution_backward(
+
+        need_gx = ctx.needs_input_grad[0]
+        gx = None
+        k = w.shape[-1]
+        # stride-1 backward-data IS a forward conv with the transposed,
+        # spatially flipped weight (dx = gy * W^T_rot) -> reuse the MFMA
+        # kernel when eligible (K*k*k % 64 == 0, same-size output)
+        if (need_gx and stride == 1 and w.shape[0] % 8 == 0
+                and (w.shape[0] * k * k) % 64 == 0
+                and x.shape[-2:] == gy.shape[-2:]):
+            from .functional import require_hip
+
+            wt = (w.transpose(0, 1).flip(-1, -2)
+                  .contiguous(memory_format=torch.channels_last))
+            gx = require_hip().conv2d_fwd(gy, wt, torch.Tensor(), 1, pad, 0)
+            need_gx = False
+
+        gx2, gw, gb = torch.ops.aten.convolution_backward(
             gy, x, w, [w.shape[0]] if has_bias else None,
             [stride, stride], [pad, pad], [1, 1], False, [0, 0], 1,
-            [ctx.needs_input_grad[0], ctx.needs_input_grad[1],
+            [need_gx, ctx.needs_input_grad[1],
              has_bias and ctx.needs_input_grad[2]],
         )
+        if gx is None:
+            gx = gx2
         return gx, gw, gb, None, None, None
 
 

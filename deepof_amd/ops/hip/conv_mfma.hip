@@ -72,8 +72,17 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
 #define LDS_A(buf) (lds_all + (buf) * BM * BK)
 #define LDS_B(buf) (lds_all + 2 * BM * BK + (buf) * BN * BK)
 
-  const int tile_m = blockIdx.x / n_tiles_n;
-  const int tile_n = blockIdx.x % n_tiles_n;
+  // T1 XCD-aware remap (bijective): consecutive tiles share A-rows /
+  // B-panels; keep them on one XCD's L2.  8 XCDs on MI355X.
+  int bid = blockIdx.x;
+  {
+    const int nwg = gridDim.x;
+    const int q = nwg / 8, rr = nwg % 8;
+    const int xcd = bid % 8, idx = bid / 8;
+    bid = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int tile_m = bid / n_tiles_n;
+  const int tile_n = bid % n_tiles_n;
   const int m0 = tile_m * BM;
   const int n0 = tile_n * BN;
   const int M = B * OH * OW;

@@ -257,3 +257,26 @@ def test_fused_conv_hip_path_used():
     m.conv.to(torch.bfloat16)
     y = m._hip(x)
     assert y is not None and y.shape == (2, 128, 32, 32)
+
+
+def test_fused_conv_stride1_backward_matches_miopen():
+    """stride-1 backward-data via the MFMA fwd kernel (transposed,
+    flipped weight) must match aten.convolution_backward."""
+    from deepof_amd.ops.conv import FusedConvAct
+
+    torch.manual_seed(0)
+    m = FusedConvAct(64, 64, 3, 1, "elu").to(DEV)
+    x = (torch.randn(2, 64, 32, 32, device=DEV, dtype=torch.bfloat16)
+         .to(memory_format=torch.channels_last).requires_grad_(True))
+    m.conv.to(torch.bfloat16)
+    y = m._hip(x)
+    g = torch.randn_like(y)
+    y.backward(g)
+    gx_ours = x.grad.clone()
+
+    x2 = x.detach().clone().requires_grad_(True)
+    y2 = torch.nn.functional.elu(torch.nn.functional.conv2d(
+        x2, m.conv.weight, m.conv.bias, stride=1, padding=1))
+    y2.backward(g)
+    torch.testing.assert_close(gx_ours.float(), x2.grad.float(),
+                               rtol=0.05, atol=0.05)
