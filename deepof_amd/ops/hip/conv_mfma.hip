@@ -43,10 +43,14 @@ __device__ inline float act_fn(float x) {
   return x;
 }
 
-// XOR swizzle on bf16 element index within a [ROW][64] tile:
-// spread each 16-lane ds_read_b128 group across 8 16-B slots.
+// XOR swizzle on bf16 element index within a [ROW][64] (128-B row)
+// tile.  A b128 fragment read group is 16 consecutive rows at one
+// 16-B column chunk; its 16-B slot index (addr bits 4-7) is
+// (chunk ^ f(row)) | ((row&1)<<3), so f(row) = (row>>1)&7 makes all
+// 16 rows land on 16 distinct slots of the 256-B bank row ->
+// conflict-free ds_read_b128 ((row&7) alone leaves r / r+8 2-way).
 __device__ inline int swz(int row, int col) {
-  return col ^ ((row & 7) << 3);
+  return col ^ (((row >> 1) & 7) << 3);
 }
 
 // ---------------------------------------------------------------------
@@ -137,10 +141,11 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
     const int rs = rsc0 / C;       // uniform: C % BK == 0
     const int r = rs / S, ss = rs % S;
     const int c0 = rsc0 % C;
-    const int col = g_colb ^ ((g_row_in_grp & 7) << 3);  // source swizzle
 #pragma unroll
     for (int gi = 0; gi < A_GROUPS / 4; ++gi) {
       const int g = wid + gi * 4;
+      const int row = g * 8 + g_row_in_grp;
+      const int col = g_colb ^ (((row >> 1) & 7) << 3);  // source swizzle
       const int iy = ga_iy[gi] + r;
       const int ix = ga_ix[gi] + ss;
       const bf16* src = zero_page;
@@ -155,7 +160,9 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
 #pragma unroll
     for (int gi = 0; gi < B_GROUPS / 4; ++gi) {
       const int g = wid + gi * 4;
-      const int n = n0 + g * 8 + g_row_in_grp;
+      const int row = g * 8 + g_row_in_grp;
+      const int col = g_colb ^ (((row >> 1) & 7) << 3);
+      const int n = n0 + row;
       const bf16* src = (n < K)
           ? w + (long)n * (R * S * C) + rsc0 + col
           : zero_page;
@@ -353,5 +360,252 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
   }
 #undef LAUNCH_ACT
 #undef LAUNCH
+  return out;
+}
+
+// =====================================================================
+// Deep-pipelined 256x256 conv kernel (8 waves, 4 phases per K-tile).
+//
+// The 128x128 2-barrier structure above tops out ~700 TF: its
+// stage -> vmcnt(0) -> barrier path serializes (cdna guide §5).  This
+// kernel applies the guide's 256² deep-pipeline recipe to the conv:
+//  - K-tile (BK=64) split into two 32-wide k-halves; A and B halves
+//    (256x32 bf16 = 16 KiB each) are staged by global_load_lds one
+//    half per phase, FOUR phases ahead of consumption,
+//  - counted s_waitcnt vmcnt(4) (never 0 mid-loop) + raw s_barrier:
+//    prefetched halves stay in flight across barriers,
+//  - s_setprio(1) around each 16-MFMA cluster (phase role split),
+//  - LDS swizzle: phys = logical ^ (((row>>2)&3)<<4) bytes -> the 16
+//    rows of a b128 fragment-read group cover 16 distinct 16-B slots
+//    (conflict-free); applied on the glds SOURCE address and the read.
+// LDS: 2 dbuf x 2 khalf x (A 16K + B 16K) = 128 KiB -> 1 block/CU.
+// =====================================================================
+namespace {
+
+template <int ACT>
+__global__ __launch_bounds__(512)
+void conv_fwd_mfma256_kernel(const bf16* __restrict__ x,
+                             const bf16* __restrict__ w,
+                             const float* __restrict__ bias,
+                             bf16* __restrict__ out,
+                             const bf16* __restrict__ zero_page,
+                             int B, int IH, int IW, int C,
+                             int K, int R, int S, int OH, int OW,
+                             int stride, int pad, int n_tiles_n) {
+  constexpr int BM = 256, BN = 256, BK = 64;
+  // one __shared__ object (hipcc trap: a second one forces vmcnt(0)
+  // before every ds_read of a glds pipeline)
+  __shared__ bf16 lds_all[2 * 2 * (256 * 32) * 2];  // dbuf x khalf x (A+B)
+#define LDS256_A(buf, kh) (lds_all + ((buf) * 2 + (kh)) * (256 * 32))
+#define LDS256_B(buf, kh) (lds_all + 4 * (256 * 32) + ((buf) * 2 + (kh)) * (256 * 32))
+
+  int bid = blockIdx.x;
+  {
+    const int nwg = gridDim.x;
+    const int q = nwg / 8, rr = nwg % 8;
+    const int xcd = bid % 8, idx = bid / 8;
+    bid = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const int tile_m = bid / n_tiles_n;
+  const int tile_n = bid % n_tiles_n;
+  const int m0 = tile_m * BM;
+  const int n0 = tile_n * BN;
+  const int M = B * OH * OW;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;              // 8 waves, 2M x 4N
+  const int wm = (wid >> 2) * 128;       // wave row offset (0 / 128)
+  const int wn = (wid & 3) * 64;         // wave col offset
+
+  // ---- staging geometry: 1-KiB glds covers 16 rows x 64 B ----
+  const int st_row_in_grp = lane >> 2;       // 0..15
+  const int st_colb = (lane & 3) * 8;        // element col (of 32)
+  // per-wave groups: {wid, wid+8} of 16 rows each
+  int sa_iy[2], sa_ix[2];
+  long sa_base[2];
+  int sb_n[2];
+#pragma unroll
+  for (int gi = 0; gi < 2; ++gi) {
+    const int g = wid + gi * 8;
+    const int row = g * 16 + st_row_in_grp;
+    const int m = m0 + row;
+    const int mm = m < M ? m : 0;
+    const int ox = mm % OW;
+    const int oy = (mm / OW) % OH;
+    const int bb = mm / (OW * OH);
+    sa_iy[gi] = oy * stride - pad;
+    sa_ix[gi] = ox * stride - pad;
+    sa_base[gi] = (m < M) ? (long)bb * IH * IW : -1;
+    sb_n[gi] = n0 + row;
+  }
+  // source swizzle col (elements) for this lane within a 32-col half
+  auto swz_col = [&](int row) {
+    return st_colb ^ (((row >> 2) & 3) << 3);
+  };
+
+  const int n_stages = (R * S * C) / BK;
+
+  // stage one half-tile (A or B) of K-tile `t` into dbuf slot t&1
+  auto stage_half = [&](int t, int kh, bool is_a) {
+    const int rsc0 = t * BK + kh * 32;
+    const int rs = rsc0 / C;
+    const int r = rs / S, ss = rs % S;
+    const int c0 = rsc0 % C;
+    const int buf = t & 1;
+#pragma unroll
+    for (int gi = 0; gi < 2; ++gi) {
+      const int g = wid + gi * 8;
+      const int row = g * 16 + st_row_in_grp;
+      const int col = swz_col(row);
+      const bf16* src = zero_page;
+      if (is_a) {
+        const int iy = sa_iy[gi] + r;
+        const int ix = sa_ix[gi] + ss;
+        if (sa_base[gi] >= 0 && iy >= 0 && iy < IH && ix >= 0 && ix < IW)
+          src = x + (sa_base[gi] + (long)iy * IW + ix) * C + c0 + col;
+      } else {
+        if (sb_n[gi] < K)
+          src = w + (long)sb_n[gi] * (R * S * C) + rsc0 + col;
+      }
+      bf16* dst = (is_a ? LDS256_A(buf, kh) : LDS256_B(buf, kh)) + g * 16 * 32;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)dst, 16, 0, 0);
+    }
+  };
+
+  f32x4 acc[8][4] = {};
+
+  // prologue: all four halves of tile 0 (order matches the loop's)
+  stage_half(0, 0, true);   // A_k0
+  stage_half(0, 0, false);  // B_k0
+  stage_half(0, 1, true);   // A_k1
+  stage_half(0, 1, false);  // B_k1
+
+  const int frag_col = (lane >> 4) * 8;  // k offset within the 32-col half
+
+  for (int t = 0; t < n_stages; ++t) {
+    const int buf = t & 1;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int kh = p >> 1;   // k-half consumed this phase
+      const int mq = p & 1;    // m-frag quadrant (frags 4mq..4mq+3)
+
+      if (p == 0 || p == 2) {
+        // own glds for this k-half landed; leaves 4 (2 half-tiles) in
+        // flight mid-loop, drains only on the last tile
+        if (t + 1 < n_stages)
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+      }
+
+      // ds-read this phase's fragments (plain loads: hipcc emits the
+      // fine-grained lgkmcnt waits before the MFMAs itself)
+      bf16x8 afrag[4], bfrag[4];
+      const bf16* a_base = LDS256_A(buf, kh);
+      const bf16* b_base = LDS256_B(buf, kh);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) {
+        const int row = wm + (mq * 4 + mf) * 16 + (lane & 15);
+        const int col = frag_col ^ (((row >> 2) & 3) << 3);
+        afrag[mf] = *reinterpret_cast<const bf16x8*>(a_base + row * 32 + col);
+      }
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int row = wn + nf * 16 + (lane & 15);
+        const int col = frag_col ^ (((row >> 2) & 3) << 3);
+        bfrag[nf] = *reinterpret_cast<const bf16x8*>(b_base + row * 32 + col);
+      }
+
+      // issue next tile's half for THIS phase slot (4 phases ahead)
+      if (t + 1 < n_stages) {
+        if (p == 0) stage_half(t + 1, 0, true);
+        else if (p == 1) stage_half(t + 1, 0, false);
+        else if (p == 2) stage_half(t + 1, 1, true);
+        else stage_half(t + 1, 1, false);
+      }
+
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mq * 4 + mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              __builtin_bit_cast(short8v, afrag[mf]),
+              __builtin_bit_cast(short8v, bfrag[nf]),
+              acc[mq * 4 + mf][nf], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // ---- epilogue ----
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int n = n0 + wn + nf * 16 + (lane & 15);
+      if (n >= K) continue;
+      const float bv = bias ? bias[n] : 0.f;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int m = m0 + wm + mf * 16 + (lane >> 4) * 4 + reg;
+        if (m >= M) continue;
+        const float val = act_fn<ACT>(acc[mf][nf][reg] + bv);
+        out[(long)m * K + n] = (bf16)val;
+      }
+    }
+  }
+#undef LDS256_A
+#undef LDS256_B
+}
+
+}  // namespace
+
+at::Tensor conv2d_fwd256(at::Tensor x, at::Tensor w, at::Tensor bias,
+                         long stride, long pad, long act) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast));
+  const int B = x.size(0), C = x.size(1), IH = x.size(2), IW = x.size(3);
+  const int K = w.size(0), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(C % 32 == 0, "conv2d_fwd256 needs C % 32 == 0");
+  TORCH_CHECK((R * S * C) % 64 == 0);
+  const int OH = (IH + 2 * (int)pad - R) / (int)stride + 1;
+  const int OW = (IW + 2 * (int)pad - S) / (int)stride + 1;
+  const long M = (long)B * OH * OW;
+
+  auto out = at::empty({B, K, OH, OW},
+                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  const float* bptr = nullptr;
+  at::Tensor bias_f;
+  if (bias.defined() && bias.numel()) {
+    bias_f = bias.to(at::kFloat).contiguous();
+    bptr = bias_f.data_ptr<float>();
+  }
+  static at::Tensor zero_page;
+  if (!zero_page.defined() || zero_page.device() != x.device())
+    zero_page = at::zeros({64}, x.options());
+
+  const bf16* xp = reinterpret_cast<const bf16*>(x.data_ptr());
+  const bf16* wp = reinterpret_cast<const bf16*>(w.data_ptr());
+  bf16* op = reinterpret_cast<bf16*>(out.data_ptr());
+  const bf16* zp = reinterpret_cast<const bf16*>(zero_page.data_ptr());
+
+  const int n_tiles_n = (K + 255) / 256;
+  const long n_blocks = ((M + 255) / 256) * n_tiles_n;
+  const dim3 grid((unsigned)n_blocks), block(512);
+#define L256(ACT_) \
+  hipLaunchKernelGGL((conv_fwd_mfma256_kernel<ACT_>), grid, block, 0,     \
+                     deepof_stream(), xp, wp, bptr, op, zp, B, IH, IW, C, \
+                     K, R, S, OH, OW, (int)stride, (int)pad, n_tiles_n)
+  if (act == 1) L256(1);
+  else if (act == 2) L256(2);
+  else if (act == 3) L256(3);
+  else L256(0);
+#undef L256
   return out;
 }

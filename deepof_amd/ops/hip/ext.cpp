@@ -21,6 +21,8 @@ std::vector<at::Tensor> correlation_backward(at::Tensor gout, at::Tensor f1,
                                              at::Tensor f2, long md);
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                       long stride, long pad, long act);
+at::Tensor conv2d_fwd256(at::Tensor x, at::Tensor w, at::Tensor bias,
+                         long stride, long pad, long act);
 at::Tensor build_adam_table(std::vector<at::Tensor> params,
                             std::vector<at::Tensor> grads,
                             std::vector<at::Tensor> exp_avgs,
@@ -50,4 +52,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam_table", &fused_adam_table, "Adam step from table");
   m.def("conv2d_fwd", &conv2d_fwd,
         "MFMA implicit-GEMM conv + bias + act (NHWC bf16)");
+  m.def("conv2d_fwd256", &conv2d_fwd256,
+        "deep-pipelined 256x256 MFMA conv (counted vmcnt, raw barriers)");
 }

@@ -47,8 +47,8 @@ def main():
     hip = require_hip()
     torch.backends.cudnn.benchmark = True
     dev = "cuda:0"
-    print(f"{'layer':9s} {'ours ms':>9s} {'miopen ms':>10s} {'speedup':>8s} "
-          f"{'TF/s':>7s}  maxerr")
+    print(f"{'layer':9s} {'o128 ms':>9s} {'o256 ms':>9s} {'miopen ms':>10s} "
+          f"{'speedup':>8s} {'TF/s':>7s}  maxerr")
     for name, B, C, H, W, K, R, stride in SHAPES:
         pad = R // 2
         torch.manual_seed(0)
@@ -65,6 +65,11 @@ def main():
         def ours():
             return hip.conv2d_fwd(x, w, b, stride, pad, 1)
 
+        def ours256():
+            return hip.conv2d_fwd256(x, w, b, stride, pad, 1)
+
+        use256 = C % 32 == 0 and K >= 192
+
         try:
             y1 = ours()
         except Exception as e:
@@ -76,14 +81,22 @@ def main():
                              padding=pad))
         err_ours = (y1.float() - ref).abs().max().item()
         err_mio = (y0.float() - ref).abs().max().item()
+        err_256 = float('nan')
+        t_256 = float('nan')
+        if use256:
+            y2 = ours256()
+            err_256 = (y2.float() - ref).abs().max().item()
+            t_256 = bench(ours256)
         t_ours = bench(ours)
         t_mio = bench(miopen)
         oh = (H + 2 * pad - R) // stride + 1
         ow = (W + 2 * pad - R) // stride + 1
         flops = 2.0 * B * oh * ow * K * C * R * R
-        tf = flops / (t_ours / 1000) / 1e12
-        print(f"{name:9s} {t_ours:9.3f} {t_mio:10.3f} {t_mio/t_ours:8.2f} "
-              f"{tf:7.1f}  ours={err_ours:.3e} miopen={err_mio:.3e}")
+        tbest = min(t_ours, t_256) if use256 else t_ours
+        tf = flops / (tbest / 1000) / 1e12
+        print(f"{name:9s} {t_ours:9.3f} {t_256:9.3f} {t_mio:10.3f} "
+              f"{t_mio/tbest:8.2f} {tf:7.1f}  ours={err_ours:.3e} "
+              f"o256={err_256:.3e} miopen={err_mio:.3e}")
 
 
 if __name__ == "__main__":
