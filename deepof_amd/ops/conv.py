@@ -34,12 +34,13 @@ def _act(y, act: str | None):
 
 class _FusedConvFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, bias, stride, pad, act_code):
+    def forward(ctx, x, w, bias, stride, pad, act_code, variant="hip128"):
         from .functional import require_hip
 
-        y = require_hip().conv2d_fwd(
-            x, w, bias if bias is not None else torch.Tensor(),
-            stride, pad, act_code)
+        hip = require_hip()
+        fn = hip.conv2d_fwd256 if variant == "hip256" else hip.conv2d_fwd
+        y = fn(x, w, bias if bias is not None else torch.Tensor(),
+               stride, pad, act_code)
         ctx.save_for_backward(x, w, y)
         ctx.meta = (stride, pad, act_code, bias is not None)
         return y
@@ -81,7 +82,7 @@ class _FusedConvFn(torch.autograd.Function):
         )
         if gx is None:
             gx = gx2
-        return gx, gw, gb, None, None, None
+        return gx, gw, gb, None, None, None, None
 
 
 class FusedConvAct(nn.Module):
@@ -106,7 +107,7 @@ class FusedConvAct(nn.Module):
     def _miopen(self, x):
         return _act(self.conv(x), self.act_name)
 
-    def _hip(self, x):
+    def _hip(self, x, variant="hip128"):
         w = self.conv.weight
         if torch.is_autocast_enabled():
             x = x.to(torch.bfloat16)
@@ -116,7 +117,7 @@ class FusedConvAct(nn.Module):
         w = w.contiguous(memory_format=torch.channels_last)
         x = x.contiguous(memory_format=torch.channels_last)
         return _FusedConvFn.apply(x, w, self.conv.bias, self.stride,
-                                  self.k // 2, self.act_code)
+                                  self.k // 2, self.act_code, variant)
 
     def forward(self, x):
         if not self._hip_eligible(x):
@@ -126,8 +127,8 @@ class FusedConvAct(nn.Module):
         choice = _dispatch_cache.get(key)
         if choice is None:
             choice = self._autotune(x, key)
-        if choice == "hip":
-            y = self._hip(x)
+        if choice.startswith("hip"):
+            y = self._hip(x, choice)
             if y is not None:
                 return y
         return self._miopen(x)
@@ -150,8 +151,16 @@ class FusedConvAct(nn.Module):
             torch.cuda.synchronize()
             return time.perf_counter() - t0
 
-        t_hip = timeit(lambda: self._hip(x))
-        t_mio = timeit(lambda: self._miopen(x))
-        choice = "hip" if t_hip <= t_mio else "miopen"
+        cands = {"hip128": lambda: self._hip(x, "hip128"),
+                 "miopen": lambda: self._miopen(x)}
+        if self.cin % 32 == 0 and self.conv.out_channels >= 192:
+            cands["hip256"] = lambda: self._hip(x, "hip256")
+        times = {}
+        for name, fn in cands.items():
+            try:
+                times[name] = timeit(fn)
+            except Exception:
+                pass
+        choice = min(times, key=times.get)
         _dispatch_cache[key] = choice
         return choice
