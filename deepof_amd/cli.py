@@ -27,7 +27,18 @@ def main(argv=None):
     p_eval = sub.add_parser("eval", help="evaluate AEE from a checkpoint")
     p_eval.add_argument("--config", type=str, default=None)
     p_eval.add_argument("--checkpoint", type=str, required=False)
+    p_eval.add_argument("--dump-dir", type=str, default=None,
+                        help="write flow color maps / .flo / warped frames")
     p_eval.add_argument("overrides", nargs="*")
+
+    p_infer = sub.add_parser("infer", help="predict flow for an image pair")
+    p_infer.add_argument("--config", type=str, default=None)
+    p_infer.add_argument("--checkpoint", type=str, required=True)
+    p_infer.add_argument("--img1", type=str, required=True)
+    p_infer.add_argument("--img2", type=str, required=True)
+    p_infer.add_argument("--out", type=str, default="flow_out",
+                         help="output prefix (.flo and .jpg written)")
+    p_infer.add_argument("overrides", nargs="*")
 
     args = parser.parse_args(argv)
     cfg = Config.from_yaml(args.config) if args.config else Config()
@@ -59,8 +70,39 @@ def main(argv=None):
                                   drop_last=False, num_workers=2)
         mean = DATASET_MEANS.get(cfg.dataset, (127.5, 127.5, 127.5))
         aee = evaluate_aee(model, loader, mean, flow_scales[0], device,
-                           cfg.dataset)
+                           cfg.dataset, dump_dir=args.dump_dir)
         print(f"AEE: {aee:.4f}")
+    elif args.command == "infer":
+        import numpy as np
+        import torch
+        from PIL import Image
+
+        from .data.image import load_image, to_chw
+        from .engine.evaluator import predict_flow
+        from .losses.unsup import DATASET_MEANS
+        from .models import build_model
+        from .utils import flow_to_color, write_flo
+
+        model, flow_scales, _ = build_model(cfg.model, act=cfg.activation)
+        device = torch.device("cuda" if cfg.device == "cuda"
+                              and torch.cuda.is_available() else "cpu")
+        model.to(device).eval()
+        state = torch.load(args.checkpoint, map_location=device,
+                           weights_only=False)
+        model.load_state_dict(state["model"] if "model" in state else state)
+
+        img1 = torch.from_numpy(to_chw(load_image(args.img1))).unsqueeze(0)
+        img2 = torch.from_numpy(to_chw(load_image(args.img2))).unsqueeze(0)
+        mean = DATASET_MEANS.get(cfg.dataset, (127.5, 127.5, 127.5))
+        with torch.no_grad():
+            pred = predict_flow(model, img1.to(device), img2.to(device),
+                                mean, flow_scales[0], cfg.dataset,
+                                gt_size=tuple(img1.shape[-2:]))
+        flow = pred[0].permute(1, 2, 0).cpu().numpy()
+        write_flo(args.out + ".flo", flow)
+        Image.fromarray(flow_to_color(flow)).save(args.out + ".jpg")
+        print(f"wrote {args.out}.flo and {args.out}.jpg "
+              f"(|f| max {np.abs(flow).max():.2f})")
     return 0
 
 
