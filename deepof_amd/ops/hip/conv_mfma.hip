@@ -382,7 +382,14 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
 // =====================================================================
 namespace {
 
-template <int ACT>
+template <int SWZV>
+__device__ inline int swz256f(int row) {
+  if (SWZV == 1) return (row >> 1) & 3;
+  if (SWZV == 2) return ((row >> 1) & 3) ^ ((row >> 3) & 3);
+  return (row >> 2) & 3;
+}
+
+template <int ACT, int SWZV = 0>
 __global__ __launch_bounds__(512)
 void conv_fwd_mfma256_kernel(const bf16* __restrict__ x,
                              const bf16* __restrict__ w,
@@ -441,7 +448,7 @@ void conv_fwd_mfma256_kernel(const bf16* __restrict__ x,
   }
   // source swizzle col (elements) for this lane within a 32-col half
   auto swz_col = [&](int row) {
-    return st_colb ^ (((row >> 2) & 3) << 3);
+    return st_colb ^ (swz256f<SWZV>(row) << 3);
   };
 
   const int n_stages = (R * S * C) / BK;
@@ -510,13 +517,13 @@ void conv_fwd_mfma256_kernel(const bf16* __restrict__ x,
 #pragma unroll
       for (int mf = 0; mf < 4; ++mf) {
         const int row = wm + (mq * 4 + mf) * 16 + (lane & 15);
-        const int col = frag_col ^ (((row >> 2) & 3) << 3);
+        const int col = frag_col ^ (swz256f<SWZV>(row) << 3);
         afrag[mf] = *reinterpret_cast<const bf16x8*>(a_base + row * 32 + col);
       }
 #pragma unroll
       for (int nf = 0; nf < 4; ++nf) {
         const int row = wn + nf * 16 + (lane & 15);
-        const int col = frag_col ^ (((row >> 2) & 3) << 3);
+        const int col = frag_col ^ (swz256f<SWZV>(row) << 3);
         bfrag[nf] = *reinterpret_cast<const bf16x8*>(b_base + row * 32 + col);
       }
 
@@ -598,14 +605,25 @@ at::Tensor conv2d_fwd256(at::Tensor x, at::Tensor w, at::Tensor bias,
   const int n_tiles_n = (K + 255) / 256;
   const long n_blocks = ((M + 255) / 256) * n_tiles_n;
   const dim3 grid((unsigned)n_blocks), block(512);
-#define L256(ACT_) \
-  hipLaunchKernelGGL((conv_fwd_mfma256_kernel<ACT_>), grid, block, 0,     \
-                     deepof_stream(), xp, wp, bptr, op, zp, B, IH, IW, C, \
-                     K, R, S, OH, OW, (int)stride, (int)pad, n_tiles_n)
-  if (act == 1) L256(1);
-  else if (act == 2) L256(2);
-  else if (act == 3) L256(3);
-  else L256(0);
+  static int swzv = [] {
+    const char* e = getenv("DEEPOF_CONV256_SWZ");
+    return e ? atoi(e) : 0;
+  }();
+#define L256(ACT_, SWZ_) \
+  hipLaunchKernelGGL((conv_fwd_mfma256_kernel<ACT_, SWZ_>), grid, block,  \
+                     0, deepof_stream(), xp, wp, bptr, op, zp, B, IH, IW, \
+                     C, K, R, S, OH, OW, (int)stride, (int)pad, n_tiles_n)
+#define L256A(SWZ_)                              \
+  do {                                           \
+    if (act == 1) L256(1, SWZ_);                 \
+    else if (act == 2) L256(2, SWZ_);            \
+    else if (act == 3) L256(3, SWZ_);            \
+    else L256(0, SWZ_);                          \
+  } while (0)
+  if (swzv == 1) L256A(1);
+  else if (swzv == 2) L256A(2);
+  else L256A(0);
+#undef L256A
 #undef L256
   return out;
 }
