@@ -605,9 +605,11 @@ at::Tensor conv2d_fwd256(at::Tensor x, at::Tensor w, at::Tensor bias,
   const int n_tiles_n = (K + 255) / 256;
   const long n_blocks = ((M + 255) / 256) * n_tiles_n;
   const dim3 grid((unsigned)n_blocks), block(512);
+  // measured: SWZ variant 1 ((row>>1)&3) is fastest on gfx950
+  // (profiles/r01_conv_mfma_pmc.md sweep)
   static int swzv = [] {
     const char* e = getenv("DEEPOF_CONV256_SWZ");
-    return e ? atoi(e) : 0;
+    return e ? atoi(e) : 1;
   }();
 #define L256(ACT_, SWZ_) \
   hipLaunchKernelGGL((conv_fwd_mfma256_kernel<ACT_, SWZ_>), grid, block,  \
