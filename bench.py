@@ -177,7 +177,7 @@ def main():
     imgs_per_sec = args.batch * world * args.steps / elapsed
     if rank == 0:
         print(json.dumps({
-            "metric": "imgs/sec FlowNetS@FlyingChairs",
+            "metric": f"imgs/sec {args.model}@FlyingChairs",
             "value": imgs_per_sec,
             "unit": "imgs/sec",
             "n_gpus": world,

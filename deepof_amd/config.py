@@ -47,6 +47,7 @@ class Config:
     guided: bool = False                 # add proxy-label supervision
     guided_weight: float = 1.0
     photo_weight: float = 1.0
+    perceptual_weight: float = 0.0       # VGG feature-space warp loss
 
     # optimization
     lr: float = 1.6e-5
@@ -69,6 +70,7 @@ class Config:
     resume: bool = True
 
     nan_restart_limit: int = 3           # auto-restarts from ckpt on NaN
+    profile_steps: int = 0               # torch.profiler trace of N steps
 
     # action head (UCF101 joint training)
     action_classes: int = 0              # >0 enables the action head

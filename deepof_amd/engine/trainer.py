@@ -91,6 +91,11 @@ class Trainer:
             MultiScaleGuidedLoss(self.flow_scales, self.loss_weights)
             if cfg.guided else None
         )
+        self.perceptual_loss = None
+        if cfg.perceptual_weight > 0:
+            from ..losses import PerceptualWarpLoss
+
+            self.perceptual_loss = PerceptualWarpLoss().to(self.device)
 
         if self.world > 1:
             self.model = BucketedDataParallel(self.model)
@@ -223,6 +228,20 @@ class Trainer:
                 g = self.guided_loss(flows, gt)
                 total = total + cfg.guided_weight * g["total"]
                 parts["guided"] = float(g["total"].detach())
+            if self.perceptual_loss is not None:
+                import torch.nn.functional as TF
+
+                H, W = geo1.shape[-2:]
+                flow_full = TF.interpolate(
+                    flows[0].float(), size=(H, W), mode="bilinear",
+                    align_corners=False) * (self.flow_scales[0] * 2.0)
+                from ..losses.unsup import preprocess_images as _pp
+
+                pl = self.perceptual_loss(flow_full,
+                                          _pp(geo1, self.mean_bgr),
+                                          _pp(geo2, self.mean_bgr))
+                total = total + cfg.perceptual_weight * pl
+                parts["perceptual"] = float(pl.detach())
             if logits is not None and "label" in batch:
                 labels = batch["label"].to(self.device, non_blocking=True)
                 ce = torch.nn.functional.cross_entropy(logits.float(), labels)
@@ -267,12 +286,28 @@ class Trainer:
         max_epochs = max_epochs or cfg.max_epochs
         steps_done = 0
         nan_restarts = 0
+        profiler = None
+        if cfg.profile_steps > 0 and self.rank == 0:
+            profiler = torch.profiler.profile(
+                activities=[torch.profiler.ProfilerActivity.CPU,
+                            torch.profiler.ProfilerActivity.CUDA],
+                schedule=torch.profiler.schedule(
+                    wait=1, warmup=2, active=cfg.profile_steps, repeat=1),
+                on_trace_ready=torch.profiler.tensorboard_trace_handler(
+                    self.run_dir),
+            )
+            profiler.start()
         while self.epoch < max_epochs:
             if hasattr(loader, "sampler") and hasattr(loader.sampler, "set_epoch"):
                 loader.sampler.set_epoch(self.epoch)
             t0 = time.time()
             n_imgs = 0
-            for i, batch in enumerate(loader):
+            batches = loader
+            if self.device.type == "cuda":
+                from ..data.loader import CudaPrefetcher
+
+                batches = CudaPrefetcher(loader, self.device)
+            for i, batch in enumerate(batches):
                 try:
                     parts = self.train_step(batch)
                 except FloatingPointError as e:
@@ -299,6 +334,11 @@ class Trainer:
                     self.log_metrics(rec)
                     if self.rank == 0:
                         print(f"[deepof] {rec}")
+                if profiler is not None:
+                    profiler.step()
+                    if steps_done >= cfg.profile_steps + 4:
+                        profiler.stop()
+                        profiler = None
                 if max_steps is not None and steps_done >= max_steps:
                     self.save_checkpoint()
                     return

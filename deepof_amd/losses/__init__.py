@@ -1,6 +1,8 @@
 from .unsup import (MultiFrameUnsupLoss, MultiScaleUnsupLoss,
                     preprocess_images)
 from .guided import MultiScaleGuidedLoss
+from .perceptual import PerceptualWarpLoss
 
 __all__ = ["MultiScaleUnsupLoss", "MultiFrameUnsupLoss",
-           "MultiScaleGuidedLoss", "preprocess_images"]
+           "MultiScaleGuidedLoss", "PerceptualWarpLoss",
+           "preprocess_images"]

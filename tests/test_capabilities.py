@@ -171,3 +171,38 @@ def test_trainer_volume_step(tmp_path):
     vol = torch.rand(1, 9, 64, 96) * 255
     parts = tr.train_step({"volume": vol})
     assert np.isfinite(parts["total"])
+
+
+def test_perceptual_warp_loss():
+    from deepof_amd.losses import PerceptualWarpLoss
+
+    torch.manual_seed(0)
+    fn = PerceptualWarpLoss(levels=(3, 4))  # p2, p1 (cheap on CPU)
+    img1 = torch.rand(1, 3, 64, 64)
+    img2 = torch.rand(1, 3, 64, 64)
+    flow = torch.zeros(1, 2, 64, 64, requires_grad=True)
+    loss = fn(flow, img1, img2)
+    assert torch.isfinite(loss)
+    loss.backward()
+    assert torch.isfinite(flow.grad).all()
+    # identical images + zero flow -> near-minimal loss vs random flow
+    loss0 = fn(torch.zeros(1, 2, 64, 64), img1, img1)
+    lossr = fn(torch.randn(1, 2, 64, 64) * 8, img1, img1)
+    assert float(loss0) < float(lossr)
+
+
+def test_trainer_profiler_smoke(tmp_path):
+    import os
+
+    from deepof_amd.config import Config
+    from deepof_amd.engine import Trainer
+
+    cfg = Config.from_dict(dict(
+        dataset="synthetic", image_size=(32, 48), batch_size=2,
+        num_workers=0, model="flownets", precision="fp32", device="cpu",
+        log_dir=str(tmp_path), run_name="p", profile_steps=2,
+        log_interval=100,
+    ))
+    Trainer(cfg).fit(max_steps=6)
+    files = os.listdir(os.path.join(str(tmp_path), "p"))
+    assert any("trace" in f or f.endswith(".json") for f in files), files
