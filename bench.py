@@ -52,10 +52,16 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    # CPU/gloo fallback exists so the FULL distributed path (DDP wrapper,
+    # barriers, reduction, JSON contract) is testable without a GPU
+    use_cuda = torch.cuda.is_available()
     if world > 1:
-        dist.init_process_group("nccl")
-    torch.cuda.set_device(local_rank)
-    dev = torch.device(f"cuda:{local_rank}")
+        dist.init_process_group("nccl" if use_cuda else "gloo")
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+        dev = torch.device(f"cuda:{local_rank}")
+    else:
+        dev = torch.device("cpu")
     torch.manual_seed(1234 + rank)
 
     torch.backends.cudnn.benchmark = True  # MIOpen find for each conv shape
@@ -84,7 +90,7 @@ def main():
             x = x.to(memory_format=torch.channels_last)
         batches.append((x, img1, img2))
 
-    use_bf16 = args.dtype == "bf16"
+    use_bf16 = args.dtype == "bf16" and use_cuda
 
     def step(i):
         x, img1, img2 = batches[i % n_batches]
@@ -115,7 +121,7 @@ def main():
         opt._hyper_pin[1] = 1.0 - b1**nstep
         opt._hyper_pin[2] = 1.0 - b2**nstep
 
-    if args.graphs and world == 1:
+    if args.graphs and world == 1 and use_cuda:
         try:
             for i in range(max(args.warmup, 3)):
                 step(i)
@@ -157,12 +163,14 @@ def main():
 
     if world > 1:
         dist.barrier()
-    torch.cuda.synchronize()
+    if use_cuda:
+        torch.cuda.synchronize()
     t0 = time.perf_counter()
     last = None
     for i in range(args.steps):
         last = timed_step(i)
-    torch.cuda.synchronize()
+    if use_cuda:
+        torch.cuda.synchronize()
     if world > 1:
         dist.barrier()
     t1 = time.perf_counter()

@@ -108,8 +108,17 @@ class BucketedDataParallel(torch.nn.Module):
             off = 0
             for p in g:
                 n = p.numel()
-                # autograd accumulates straight into the flat slice
-                p.grad = flat[off : off + n].view_as(p)
+                # autograd accumulates straight into the flat slice; the
+                # view's strides must match the param's memory format or
+                # AccumulateGrad copies every step (channels_last convs)
+                sl = flat[off : off + n]
+                if (p.dim() == 4 and
+                        p.is_contiguous(memory_format=torch.channels_last)
+                        and not p.is_contiguous()):
+                    N, C, H, W = p.shape
+                    p.grad = sl.view(N, H, W, C).permute(0, 3, 1, 2)
+                else:
+                    p.grad = sl.view_as(p)
                 off += n
                 self._param_bucket[id(p)] = bucket
                 h = p.register_post_accumulate_grad_hook(self._on_grad_ready)
