@@ -97,8 +97,8 @@ class FlowNetC(nn.Module):
         c1b = self.conv1(im2)
         c2b = self.conv2(c1b)
         c3b = self.conv3(c2b)
-        # correlation in fp32 (bf16 dot over 256 channels loses too much)
-        corr = correlation(c3a.float(), c3b.float(), self.md).to(c3a.dtype)
+        # bf16 inputs, fp32 accumulation inside the HIP kernel
+        corr = correlation(c3a, c3b, self.md).to(c3a.dtype)
         redir = self.conv_redir(c3a)
         c3 = self.conv3_1(torch.cat([corr, redir], dim=1))
         c4 = self.conv4_2(self.conv4_1(c3))

@@ -136,9 +136,11 @@ def test_correlation_matches_reference():
     f1 = torch.randn(2, 16, 20, 28)
     f2 = torch.randn(2, 16, 20, 28)
     md = 3
-    want = ref.correlation(f1, f2, md)
+    # the kernel stages bf16 (fp32 accumulation): compare against the
+    # reference on identically-rounded inputs
+    want = ref.correlation(f1.bfloat16().float(), f2.bfloat16().float(), md)
     got = _hip().correlation_forward(f1.to(DEV), f2.to(DEV), md).cpu()
-    torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(got, want, rtol=2e-3, atol=2e-3)
 
 
 def test_correlation_backward_matches_autograd():
@@ -146,13 +148,13 @@ def test_correlation_backward_matches_autograd():
     f1 = torch.randn(1, 8, 10, 12, requires_grad=True)
     f2 = torch.randn(1, 8, 10, 12, requires_grad=True)
     md = 2
-    out = ref.correlation(f1, f2, md)
+    out = ref.correlation(f1.bfloat16().float(), f2.bfloat16().float(), md)
     g = torch.randn_like(out)
     out.backward(g)
     g1, g2 = _hip().correlation_backward(g.to(DEV), f1.detach().to(DEV),
                                          f2.detach().to(DEV), md)
-    torch.testing.assert_close(g1.cpu(), f1.grad, rtol=1e-4, atol=1e-5)
-    torch.testing.assert_close(g2.cpu(), f2.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(g1.cpu(), f1.grad, rtol=2e-3, atol=2e-3)
+    torch.testing.assert_close(g2.cpu(), f2.grad, rtol=2e-3, atol=2e-3)
 
 
 def test_fused_adam_matches_cpu():
