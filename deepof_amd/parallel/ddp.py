@@ -80,8 +80,14 @@ class BucketedDataParallel(torch.nn.Module):
     # -- setup ------------------------------------------------------------
     def _broadcast_params(self):
         for t in self.module.state_dict().values():
-            if isinstance(t, torch.Tensor) and t.numel():
+            if not (isinstance(t, torch.Tensor) and t.numel()):
+                continue
+            if t.is_contiguous():
                 dist.broadcast(t.data, src=0, group=self.pg)
+            else:  # channels_last params: collectives want contiguous
+                buf = t.data.contiguous()
+                dist.broadcast(buf, src=0, group=self.pg)
+                t.data.copy_(buf)
 
     def _build_buckets(self, cap_bytes: int):
         params = [p for p in self.module.parameters() if p.requires_grad]

@@ -31,9 +31,10 @@ def _worker(rank, world, port, results):
     try:
         from deepof_amd.parallel import BucketedDataParallel
 
-        model = BucketedDataParallel(_make_model(), bucket_cap_mb=4)
+        base = _make_model().to(memory_format=torch.channels_last)
+        model = BucketedDataParallel(base, bucket_cap_mb=4)
         torch.manual_seed(100 + rank)
-        x = torch.randn(2, 6, 32, 48)
+        x = torch.randn(2, 6, 32, 48).to(memory_format=torch.channels_last)
         loss = _loss(model, x)
         loss.backward()
         model.finish_gradient_sync()
@@ -62,11 +63,12 @@ def test_bucketed_ddp_matches_single_process():
         ddp_grads = {k: v for k, v in results[0].items()}
 
     # single-process on the combined batch: average of per-rank losses
-    model = _make_model()
+    model = _make_model().to(memory_format=torch.channels_last)
     xs = []
     for rank in range(world):
         torch.manual_seed(100 + rank)
-        xs.append(torch.randn(2, 6, 32, 48))
+        xs.append(torch.randn(2, 6, 32, 48).to(
+            memory_format=torch.channels_last))
     loss = sum(_loss(model, x) for x in xs) / world
     loss.backward()
 
