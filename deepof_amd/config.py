@@ -43,6 +43,7 @@ class Config:
     alpha_c: float = 0.25
     alpha_s: float = 0.37
     lambda_smooth: float = 1.0
+    edge_aware: bool = False             # loss_interp_bk image-gradient masks
     loss_weights: Optional[list] = None  # default from model registry
     guided: bool = False                 # add proxy-label supervision
     guided_weight: float = 1.0

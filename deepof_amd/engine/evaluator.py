@@ -89,3 +89,27 @@ def _dump_artifacts(dump_dir, batch_idx, img1, img2, pred, gt):
     rec = recon[0].permute(1, 2, 0).cpu().numpy()[:, :, ::-1]  # BGR->RGB
     Image.fromarray(np.clip(rec, 0, 255).astype(np.uint8)).save(
         os.path.join(dump_dir, f"b{batch_idx:04d}_warped.jpg"))
+
+
+@torch.no_grad()
+def evaluate_accuracy(model, loader, mean_bgr, device,
+                      max_batches=None) -> float:
+    """UCF101 action accuracy (parity: ucf101train.py:183, :273)."""
+    from ..losses.unsup import preprocess_images
+
+    model.eval()
+    correct = total = 0
+    for i, batch in enumerate(loader):
+        if max_batches is not None and i >= max_batches:
+            break
+        img1 = batch["img1"].to(device, non_blocking=True)
+        img2 = batch["img2"].to(device, non_blocking=True)
+        labels = batch["label"].to(device, non_blocking=True)
+        x = torch.cat([preprocess_images(img1.float(), mean_bgr),
+                       preprocess_images(img2.float(), mean_bgr)], dim=1)
+        out = model(x)
+        logits = out[1] if isinstance(out, tuple) else out
+        correct += int((logits.argmax(1) == labels).sum())
+        total += labels.numel()
+    model.train()
+    return correct / max(total, 1)

@@ -86,6 +86,7 @@ class Trainer:
         self.unsup_loss = MultiScaleUnsupLoss(
             self.flow_scales, self.loss_weights, mean,
             cfg.epsilon, cfg.alpha_c, cfg.alpha_s, cfg.lambda_smooth,
+            edge_aware=cfg.edge_aware,
         )
         self.guided_loss = (
             MultiScaleGuidedLoss(self.flow_scales, self.loss_weights)

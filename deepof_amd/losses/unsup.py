@@ -44,9 +44,11 @@ class MultiScaleUnsupLoss(nn.Module):
         alpha_c: float = 0.25,
         alpha_s: float = 0.37,
         lambda_smooth: float = 1.0,
+        edge_aware: bool = False,
     ):
         super().__init__()
         assert len(flow_scales) == len(loss_weights)
+        self.edge_aware = edge_aware
         self.flow_scales = flow_scales
         self.loss_weights = loss_weights
         self.mean_bgr = mean_bgr
@@ -81,11 +83,23 @@ class MultiScaleUnsupLoss(nn.Module):
         scale_losses = []
         recon = None
         for k, flow in enumerate(flows):
-            res = ops.unsup_loss_scale(
-                flow.float(), pyr1[k], pyr2[k], self.flow_scales[k],
-                self.epsilon, self.alpha_c, self.alpha_s, self.lambda_smooth,
-                return_recon=(want_recon and k == 0),
-            )
+            if self.edge_aware:
+                # loss_interp_bk variant: torch-op path (image-gradient
+                # masks; the fused HIP kernel covers the default config)
+                from ..ops.reference import unsup_loss_scale_edge_aware
+
+                res = unsup_loss_scale_edge_aware(
+                    flow.float(), pyr1[k], pyr2[k], self.flow_scales[k],
+                    self.epsilon, self.alpha_c, self.alpha_s,
+                    self.lambda_smooth,
+                )
+            else:
+                res = ops.unsup_loss_scale(
+                    flow.float(), pyr1[k], pyr2[k], self.flow_scales[k],
+                    self.epsilon, self.alpha_c, self.alpha_s,
+                    self.lambda_smooth,
+                    return_recon=(want_recon and k == 0),
+                )
             if want_recon and k == 0:
                 recon = res.pop("recon")
             scale_losses.append(res)

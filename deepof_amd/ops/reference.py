@@ -303,3 +303,68 @@ def unsup_loss_scale(
     if return_recon:
         out["recon"] = recon
     return out
+
+
+# ---------------------------------------------------------------------------
+# Edge-aware smoothness weighting (loss_interp_bk / needImageGradients)
+# ---------------------------------------------------------------------------
+def image_gradient_masks(img1: torch.Tensor) -> torch.Tensor:
+    """exp-style edge weights (1 - |normalized Sobel gradient|).
+
+    Reproduces the needImageGradients branch
+    (/root/reference/version1/model/warpflow.py:92-117): per-image
+    0-255 rescale, grayscale, Sobel x/y, normalize by the max |g|,
+    weight = 1 - |g|.  Returns [B, 2, H, W] (x-weight, y-weight).
+    """
+    b = img1.shape[0]
+    mn = img1.amin(dim=(1, 2, 3), keepdim=True)
+    mx = img1.amax(dim=(1, 2, 3), keepdim=True)
+    x = (img1 - mn) / (mx - mn + 1e-12) * 255.0
+    gray = x.mean(dim=1, keepdim=True)
+    sob_x = torch.tensor([[-1.0, 0, 1], [-2, 0, 2], [-1, 0, 1]],
+                         device=img1.device).view(1, 1, 3, 3)
+    sob_y = sob_x.transpose(2, 3)
+    gx = F.conv2d(gray, sob_x, padding=1)
+    gy = F.conv2d(gray, sob_y, padding=1)
+    gx = gx / (gx.abs().amax(dim=(1, 2, 3), keepdim=True) + 1e-12)
+    gy = gy / (gy.abs().amax(dim=(1, 2, 3), keepdim=True) + 1e-12)
+    return torch.cat([1.0 - gx.abs(), 1.0 - gy.abs()], dim=1)
+
+
+def unsup_loss_scale_edge_aware(
+    flow_raw, img1, img2, flow_scale, epsilon=1e-4, alpha_c=0.25,
+    alpha_s=0.37, lambda_smooth=1.0,
+):
+    """unsup_loss_scale with edge-aware smoothness: the per-direction
+    Charbonnier terms are multiplied by the (1-|gradient|) masks before
+    the border mask (warpflow.py:147-158)."""
+    b, c, h, w = img1.shape
+    scaled = flow_raw * flow_scale
+    recon = warp_bilinear(img2, scaled)
+    bw = math.ceil(h * 0.1)
+    masked = (h - 2 * bw) > 0 and (w - 2 * bw) > 0
+    bm = border_mask(h, w, device=img1.device).to(img1.dtype) if masked else None
+    photo, num_valid = charbonnier_photometric(recon, img1, epsilon, alpha_c, bm)
+    num_valid_flows = num_valid / c * 2
+
+    dx = torch.zeros_like(scaled)
+    dy = torch.zeros_like(scaled)
+    dx[:, :, :, : w - 1] = scaled[:, :, :, : w - 1] - scaled[:, :, :, 1:]
+    dy[:, :, : h - 1, :] = scaled[:, :, : h - 1, :] - scaled[:, :, 1:, :]
+    eps2 = epsilon * epsilon
+    gmask = image_gradient_masks(img1)
+    wx = gmask[:, 0:1]
+    wy = gmask[:, 1:2]
+    ew_u = torch.pow(dx[:, 0:1] ** 2 + eps2, alpha_s) * wx + \
+        torch.pow(dy[:, 0:1] ** 2 + eps2, alpha_s) * wy
+    ew_v = torch.pow(dx[:, 1:2] ** 2 + eps2, alpha_s) * wx + \
+        torch.pow(dy[:, 1:2] ** 2 + eps2, alpha_s) * wy
+    if masked:
+        u_loss = (ew_u[:, 0] * bm).sum() / num_valid_flows
+        v_loss = (ew_v[:, 0] * bm).sum() / num_valid_flows
+    else:
+        u_loss = ew_u.mean()
+        v_loss = ew_v.mean()
+    total = photo + lambda_smooth * (u_loss + v_loss)
+    return {"total": total, "photo": photo, "u_loss": u_loss,
+            "v_loss": v_loss}

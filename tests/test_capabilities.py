@@ -206,3 +206,37 @@ def test_trainer_profiler_smoke(tmp_path):
     Trainer(cfg).fit(max_steps=6)
     files = os.listdir(os.path.join(str(tmp_path), "p"))
     assert any("trace" in f or f.endswith(".json") for f in files), files
+
+
+def test_edge_aware_smoothness():
+    from deepof_amd.ops.reference import (image_gradient_masks,
+                                          unsup_loss_scale_edge_aware)
+
+    torch.manual_seed(0)
+    img1 = torch.rand(1, 3, 24, 32)
+    img2 = torch.rand(1, 3, 24, 32)
+    gm = image_gradient_masks(img1)
+    assert gm.shape == (1, 2, 24, 32)
+    assert (gm >= -1e-5).all() and (gm <= 1.0 + 1e-5).all()
+    flow = torch.randn(1, 2, 24, 32, requires_grad=True)
+    res = unsup_loss_scale_edge_aware(flow, img1, img2, 1.0)
+    res["total"].backward()
+    assert torch.isfinite(flow.grad).all()
+    # edge-aware smoothness never exceeds the unweighted one
+    from deepof_amd.ops import reference as ref
+
+    base = ref.unsup_loss_scale(flow.detach(), img1, img2, 1.0)
+    assert float(res["u_loss"]) <= float(base["u_loss"]) + 1e-6
+
+
+def test_evaluate_accuracy():
+    from deepof_amd.data import SyntheticActionDataset, build_dataloader
+    from deepof_amd.engine.evaluator import evaluate_accuracy
+    from deepof_amd.models import STSingle
+
+    ds = SyntheticActionDataset(4, 64, 96, num_classes=5)
+    dl = build_dataloader(ds, 2, shuffle=False, num_workers=0,
+                          drop_last=False)
+    m = STSingle(input_hw=(64, 96), num_classes=5)
+    acc = evaluate_accuracy(m, dl, ds.mean_bgr, "cpu")
+    assert 0.0 <= acc <= 1.0
