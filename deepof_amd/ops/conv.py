@@ -50,13 +50,11 @@ class _FusedConvFn(torch.autograd.Function):
         x, w, y = ctx.saved_tensors
         stride, pad, act_code, has_bias = ctx.meta
         gy = gy.contiguous(memory_format=torch.channels_last)
-        if act_code == 1:  # ELU: d(pre) = go * (y>0 ? 1 : y+1)
-            gy = gy * torch.where(y > 0, torch.ones_like(y), y + 1)
-        elif act_code == 2:
-            gy = gy * torch.where(y > 0, torch.full_like(y, 1.0),
-                                  torch.full_like(y, 0.1))
-        elif act_code == 3:
-            gy = gy * (y > 0).to(gy.dtype)
+        if act_code in (1, 2, 3):
+            from .functional import require_hip
+
+            # one fused pass (gy * act_grad(y)) instead of where+mul
+            gy = require_hip().act_grad(gy, y, act_code)
 
         need_gx = ctx.needs_input_grad[0]
         gx = None

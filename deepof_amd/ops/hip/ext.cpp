@@ -16,6 +16,7 @@ at::Tensor resize_bilinear(at::Tensor x, long oh, long ow);
 at::Tensor lrn_forward(at::Tensor x, long radius, double bias, double alpha,
                        double beta);
 at::Tensor epe_sum(at::Tensor f, at::Tensor g);
+at::Tensor act_grad(at::Tensor gy, at::Tensor y, long act);
 at::Tensor correlation_forward(at::Tensor f1, at::Tensor f2, long md);
 std::vector<at::Tensor> correlation_backward(at::Tensor gout, at::Tensor f1,
                                              at::Tensor f2, long md);
@@ -45,6 +46,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("resize_bilinear", &resize_bilinear, "legacy-TF bilinear resize");
   m.def("lrn_forward", &lrn_forward, "across-channel LRN");
   m.def("epe_sum", &epe_sum, "endpoint-error sum reduction");
+  m.def("act_grad", &act_grad, "fused activation gradient from output");
   m.def("correlation_forward", &correlation_forward, "cost volume fwd");
   m.def("correlation_backward", &correlation_backward, "cost volume bwd");
   m.def("fused_adam", &fused_adam, "multi-tensor Adam step");

@@ -356,7 +356,58 @@ __global__ void epe_sum_kernel(const float* __restrict__ f,
   if (threadIdx.x == 0) atomicAdd(out, r);
 }
 
+// ---------------------------------------------------------------------
+// Fused activation gradient: gpre = gy * act'(y) from the OUTPUT y.
+// act: 1 ELU (y>0 ? 1 : y+1), 2 LeakyReLU(0.1), 3 ReLU.
+// One pass, bf16x8 vectorized (replaces torch where+mul in the fused
+// conv backward).
+// ---------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(8))) __bf16 lk_bf16x8;
+
+template <int ACT>
+__global__ void act_grad_kernel(const __bf16* __restrict__ gy,
+                                const __bf16* __restrict__ y,
+                                __bf16* __restrict__ out, long n8) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n8) return;
+  const lk_bf16x8 g = reinterpret_cast<const lk_bf16x8*>(gy)[i];
+  const lk_bf16x8 v = reinterpret_cast<const lk_bf16x8*>(y)[i];
+  lk_bf16x8 o;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const float yv = (float)v[j];
+    float d;
+    if (ACT == 1) d = yv > 0.f ? 1.f : yv + 1.f;
+    else if (ACT == 2) d = yv > 0.f ? 1.f : 0.1f;
+    else d = yv > 0.f ? 1.f : 0.f;
+    o[j] = (__bf16)((float)g[j] * d);
+  }
+  reinterpret_cast<lk_bf16x8*>(out)[i] = o;
+}
+
 }  // namespace
+
+at::Tensor act_grad(at::Tensor gy, at::Tensor y, long act) {
+  TORCH_CHECK(gy.scalar_type() == at::kBFloat16 &&
+              y.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(gy.numel() == y.numel() && gy.numel() % 8 == 0);
+  auto out = at::empty_like(gy);
+  const long n8 = gy.numel() / 8;
+  const dim3 grid((unsigned)((n8 + 255) / 256)), block(256);
+  auto* gp = reinterpret_cast<const __bf16*>(gy.data_ptr());
+  auto* yp = reinterpret_cast<const __bf16*>(y.data_ptr());
+  auto* op = reinterpret_cast<__bf16*>(out.data_ptr());
+  if (act == 1)
+    hipLaunchKernelGGL(act_grad_kernel<1>, grid, block, 0, deepof_stream(),
+                       gp, yp, op, n8);
+  else if (act == 2)
+    hipLaunchKernelGGL(act_grad_kernel<2>, grid, block, 0, deepof_stream(),
+                       gp, yp, op, n8);
+  else
+    hipLaunchKernelGGL(act_grad_kernel<3>, grid, block, 0, deepof_stream(),
+                       gp, yp, op, n8);
+  return out;
+}
 
 // =====================================================================
 // Host launchers
