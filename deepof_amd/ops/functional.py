@@ -83,8 +83,9 @@ def warp_bilinear(img2: torch.Tensor, flow: torch.Tensor) -> torch.Tensor:
 def resize_bilinear(x: torch.Tensor, out_h: int, out_w: int) -> torch.Tensor:
     if x.is_cuda:
         if x.requires_grad and torch.is_grad_enabled():
-            raise RuntimeError("resize_bilinear HIP path has no backward; "
-                               "detach the input (images are constants)")
+            # differentiable path (same legacy-TF sampling, torch ops);
+            # the HIP kernel serves the no-grad image-pyramid hot path
+            return ref.resize_bilinear(x, out_h, out_w)
         return require_hip().resize_bilinear(x.contiguous(), out_h, out_w)
     return ref.resize_bilinear(x, out_h, out_w)
 
