@@ -20,6 +20,7 @@ import torch.nn.functional as F
 
 _ACT_CODE = {None: 0, "none": 0, "elu": 1, "leaky_relu": 2, "relu": 3}
 _dispatch_cache: dict[tuple, str] = {}
+_wrw_cache: dict[tuple, str] = {}
 
 
 def _act(y, act: str | None):
@@ -72,14 +73,23 @@ class _FusedConvFn(torch.autograd.Function):
             gx = require_hip().conv2d_fwd(gy, wt, torch.Tensor(), 1, pad, 0)
             need_gx = False
 
-        gx2, gw, gb = torch.ops.aten.convolution_backward(
+        # weight grad via the MFMA wrw kernel when it measures faster
+        need_gw = ctx.needs_input_grad[1]
+        gw = None
+        if need_gw and x.shape[1] % 64 == 0 and w.shape[0] % 8 == 0:
+            gw = _maybe_hip_wrw(gy, x, w, stride, pad)
+            if gw is not None:
+                need_gw = False
+
+        gx2, gw2, gb = torch.ops.aten.convolution_backward(
             gy, x, w, [w.shape[0]] if has_bias else None,
             [stride, stride], [pad, pad], [1, 1], False, [0, 0], 1,
-            [need_gx, ctx.needs_input_grad[1],
-             has_bias and ctx.needs_input_grad[2]],
+            [need_gx, need_gw, has_bias and ctx.needs_input_grad[2]],
         )
         if gx is None:
             gx = gx2
+        if gw is None:
+            gw = gw2
         return gx, gw, gb, None, None, None, None
 
 
@@ -162,3 +172,42 @@ class FusedConvAct(nn.Module):
         choice = min(times, key=times.get)
         _dispatch_cache[key] = choice
         return choice
+
+
+def _maybe_hip_wrw(gy, x, w, stride, pad):
+    """Measured dispatch for the weight gradient: time the MFMA wrw
+    kernel against aten.convolution_backward's wrw once per shape."""
+    import time
+
+    from .functional import require_hip
+
+    key = ("wrw", tuple(x.shape), tuple(w.shape), stride)
+    choice = _wrw_cache.get(key)
+    hip = require_hip()
+    R, S = w.shape[2], w.shape[3]
+    if choice is None:
+        def ours():
+            return hip.conv2d_wrw(gy, x, R, S, stride, pad)
+
+        def mio():
+            return torch.ops.aten.convolution_backward(
+                gy, x, w, None, [stride, stride], [pad, pad], [1, 1],
+                False, [0, 0], 1, [False, True, False])[1]
+
+        def timeit(fn, n=4):
+            fn()
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(n):
+                fn()
+            torch.cuda.synchronize()
+            return time.perf_counter() - t0
+
+        with torch.no_grad():
+            t_h = timeit(ours)
+            t_m = timeit(mio)
+        choice = "hip" if t_h < t_m else "miopen"
+        _wrw_cache[key] = choice
+    if choice == "hip":
+        return hip.conv2d_wrw(gy, x, R, S, stride, pad)
+    return None

@@ -24,6 +24,8 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                       long stride, long pad, long act);
 at::Tensor conv2d_fwd256(at::Tensor x, at::Tensor w, at::Tensor bias,
                          long stride, long pad, long act);
+at::Tensor conv2d_wrw(at::Tensor gy, at::Tensor x, long R, long S,
+                      long stride, long pad);
 at::Tensor build_adam_table(std::vector<at::Tensor> params,
                             std::vector<at::Tensor> grads,
                             std::vector<at::Tensor> exp_avgs,
@@ -56,4 +58,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA implicit-GEMM conv + bias + act (NHWC bf16)");
   m.def("conv2d_fwd256", &conv2d_fwd256,
         "deep-pipelined 256x256 MFMA conv (counted vmcnt, raw barriers)");
+  m.def("conv2d_wrw", &conv2d_wrw,
+        "MFMA weight gradient (transpose-staged, pixel-split atomics)");
 }
