@@ -46,7 +46,7 @@ void conv_wrw_mfma_kernel(const bf16* __restrict__ gy,  // [B,OH,OW,K]
                           int B, int IH, int IW, int C, int K,
                           int R, int S, int OH, int OW,
                           int stride, int pad,
-                          int n_tiles_n, int pix_per_slice) {
+                          int n_tiles_n, int pix_per_slice, int use_swz) {
   // LDS: double-buffered gyT [64][BKP] + xT [64][BKP] bf16
   __shared__ bf16 lds_all[2 * 2 * 64 * BKP];
 #define LDS_GY(buf) (lds_all + (buf) * 64 * BKP)
@@ -97,7 +97,7 @@ void conv_wrw_mfma_kernel(const bf16* __restrict__ gy,  // [B,OH,OW,K]
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int row = st_c8 + j;                     // k row
-        const int col = wswz(row, pr * 2) & (BKP - 1); // pix col (pair-aligned)
+        const int col = use_swz ? (wswz(row, pr * 2) & (BKP - 1)) : pr * 2;
         unsigned v = (unsigned)__builtin_bit_cast(unsigned short, a0[j]) |
                      ((unsigned)__builtin_bit_cast(unsigned short, a1[j]) << 16);
         *reinterpret_cast<unsigned*>(&LDS_GY(buf)[row * BKP + col]) = v;
@@ -120,7 +120,7 @@ void conv_wrw_mfma_kernel(const bf16* __restrict__ gy,  // [B,OH,OW,K]
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int row = st_c8 + j;
-        const int col = wswz(row, pr * 2) & (BKP - 1);
+        const int col = use_swz ? (wswz(row, pr * 2) & (BKP - 1)) : pr * 2;
         unsigned v = (unsigned)__builtin_bit_cast(unsigned short, b0[j]) |
                      ((unsigned)__builtin_bit_cast(unsigned short, b1[j]) << 16);
         *reinterpret_cast<unsigned*>(&LDS_X(buf)[row * BKP + col]) = v;
@@ -145,14 +145,16 @@ void conv_wrw_mfma_kernel(const bf16* __restrict__ gy,  // [B,OH,OW,K]
         const int row = wm + mi * 16 + (lane & 15);
         const int col = (kk + (lane >> 4) * 8);
         afrag[mi] = *reinterpret_cast<const bf16x8*>(
-            &LDS_GY(buf)[row * BKP + (wswz(row, col) & (BKP - 1))]);
+            &LDS_GY(buf)[row * BKP +
+                         (use_swz ? (wswz(row, col) & (BKP - 1)) : col)]);
       }
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni) {
         const int row = wn + ni * 16 + (lane & 15);
         const int col = (kk + (lane >> 4) * 8);
         bfrag[ni] = *reinterpret_cast<const bf16x8*>(
-            &LDS_X(buf)[row * BKP + (wswz(row, col) & (BKP - 1))]);
+            &LDS_X(buf)[row * BKP +
+                        (use_swz ? (wswz(row, col) & (BKP - 1)) : col)]);
       }
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
@@ -214,13 +216,17 @@ at::Tensor conv2d_wrw(at::Tensor gy, at::Tensor x, long R_, long S_,
   split = (int)((M + pix_per_slice - 1) / pix_per_slice);
 
   const dim3 grid(ktiles * ntiles, split), block(256);
+  static int use_swz = [] {
+    const char* e = getenv("DEEPOF_WRW_NOSWZ");
+    return e && atoi(e) ? 0 : 1;
+  }();
   hipLaunchKernelGGL((conv_wrw_mfma_kernel<BKP>), grid, block, 0,
                      deepof_stream(),
                      reinterpret_cast<const bf16*>(gy.data_ptr()),
                      reinterpret_cast<const bf16*>(x.data_ptr()),
                      dw_f32.data_ptr<float>(), B, IH, IW, C, K, R, S,
                      OH, OW, (int)stride, (int)pad, ntiles,
-                     pix_per_slice);
+                     pix_per_slice, use_swz);
   // [K, RSC] fp32 -> [K, C, R, S] channels_last bf16
   auto dw = dw_f32.view({K, R, S, C}).permute({0, 3, 1, 2}).to(at::kBFloat16);
   return dw.contiguous(at::MemoryFormat::ChannelsLast);
