@@ -73,10 +73,17 @@ class _FusedConvFn(torch.autograd.Function):
             gx = require_hip().conv2d_fwd(gy, wt, torch.Tensor(), 1, pad, 0)
             need_gx = False
 
-        # weight grad via the MFMA wrw kernel when it measures faster
+        # weight grad via the MFMA wrw kernel when it measures faster.
+        # r01 status: correct (rel err ~3e-3, bf16-class) but the
+        # transpose-staged 64x64 tile runs 0.25-0.9x MIOpen's igemm wrw
+        # (tools/bench_wrw.py), so the candidate is opt-in until the
+        # staging uses ds_read_tr16 / a bigger tile (round-2 work).
+        import os as _os
+
         need_gw = ctx.needs_input_grad[1]
         gw = None
-        if need_gw and x.shape[1] % 64 == 0 and w.shape[0] % 8 == 0:
+        if (need_gw and _os.environ.get("DEEPOF_WRW") == "1"
+                and x.shape[1] % 64 == 0 and w.shape[0] % 8 == 0):
             gw = _maybe_hip_wrw(gy, x, w, stride, pad)
             if gw is not None:
                 need_gw = False

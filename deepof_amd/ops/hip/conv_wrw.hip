@@ -98,9 +98,10 @@ void conv_wrw_mfma_kernel(const bf16* __restrict__ gy,  // [B,OH,OW,K]
       for (int j = 0; j < 8; ++j) {
         const int row = st_c8 + j;                     // k row
         const int col = use_swz ? (wswz(row, pr * 2) & (BKP - 1)) : pr * 2;
-        unsigned v = (unsigned)__builtin_bit_cast(unsigned short, a0[j]) |
-                     ((unsigned)__builtin_bit_cast(unsigned short, a1[j]) << 16);
-        *reinterpret_cast<unsigned*>(&LDS_GY(buf)[row * BKP + col]) = v;
+        // element-typed stores (an unsigned* write here is a TBAA
+        // violation against the bf16x8 fragment reads)
+        LDS_GY(buf)[row * BKP + col] = a0[j];
+        LDS_GY(buf)[row * BKP + col + 1] = a1[j];
       }
       // x tile: rows = rsc (c within the tap)
       auto src_px = [&](int p, bf16x8& out) {
@@ -121,9 +122,8 @@ void conv_wrw_mfma_kernel(const bf16* __restrict__ gy,  // [B,OH,OW,K]
       for (int j = 0; j < 8; ++j) {
         const int row = st_c8 + j;
         const int col = use_swz ? (wswz(row, pr * 2) & (BKP - 1)) : pr * 2;
-        unsigned v = (unsigned)__builtin_bit_cast(unsigned short, b0[j]) |
-                     ((unsigned)__builtin_bit_cast(unsigned short, b1[j]) << 16);
-        *reinterpret_cast<unsigned*>(&LDS_X(buf)[row * BKP + col]) = v;
+        LDS_X(buf)[row * BKP + col] = b0[j];
+        LDS_X(buf)[row * BKP + col + 1] = b1[j];
       }
     }
   };
