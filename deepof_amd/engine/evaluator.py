@@ -31,7 +31,8 @@ def predict_flow(model, img1_raw, img2_raw, mean_bgr, flow_scale_finest,
     """Run the model and apply the eval post-processing; returns [B,2,H,W]."""
     x1 = preprocess_images(img1_raw.float(), mean_bgr)
     x2 = preprocess_images(img2_raw.float(), mean_bgr)
-    flows = model(torch.cat([x1, x2], dim=1))
+    out = model(torch.cat([x1, x2], dim=1))
+    flows = out[0] if isinstance(out, tuple) else out  # joint models
     mult, cmin, cmax = EVAL_POSTPROC.get(dataset, EVAL_POSTPROC["flying_chairs"])
     pred = flows[0].float() * flow_scale_finest * mult
     pred = pred.clamp(cmin, cmax)

@@ -64,8 +64,12 @@ class FusedAdam(Optimizer):
                 from ..ops.functional import require_hip
 
                 hip = require_hip()
+                # moment pointers are part of the signature:
+                # load_state_dict() replaces exp_avg/exp_avg_sq tensors,
+                # which must invalidate the cached device chunk table
                 sig = (len(params), params[0].data_ptr(),
-                       grads[0].data_ptr(), params[-1].data_ptr())
+                       grads[0].data_ptr(), params[-1].data_ptr(),
+                       exp_avgs[0].data_ptr(), exp_avg_sqs[-1].data_ptr())
                 if self._table is None or self._table_sig != sig:
                     self._table = hip.build_adam_table(
                         params, grads, exp_avgs, exp_avg_sqs)

@@ -240,3 +240,15 @@ def test_evaluate_accuracy():
     m = STSingle(input_hw=(64, 96), num_classes=5)
     acc = evaluate_accuracy(m, dl, ds.mean_bgr, "cpu")
     assert 0.0 <= acc <= 1.0
+
+
+def test_predict_flow_joint_model():
+    from deepof_amd.engine.evaluator import predict_flow
+    from deepof_amd.models import STSingle
+
+    m = STSingle(input_hw=(64, 96), num_classes=5)
+    img1 = torch.rand(1, 3, 64, 96) * 255
+    img2 = torch.rand(1, 3, 64, 96) * 255
+    pred = predict_flow(m, img1, img2, (104.0, 117.0, 123.0), 10.0,
+                        "ucf101", gt_size=(64, 96))
+    assert pred.shape == (1, 2, 64, 96)
