@@ -55,11 +55,29 @@ def evaluate_aee(model, loader, mean_bgr, flow_scale_finest, device,
     for i, batch in enumerate(loader):
         if max_batches is not None and i >= max_batches:
             break
-        img1 = batch["img1"].to(device, non_blocking=True)
-        img2 = batch["img2"].to(device, non_blocking=True)
         gt = batch["flow"].to(device, non_blocking=True).float()
-        pred = predict_flow(model, img1, img2, mean_bgr, flow_scale_finest,
-                            dataset, gt_size=tuple(gt.shape[-2:]))
+        if "volume" in batch:  # Sintel multi-frame: score the first pair
+            vol = batch["volume"].to(device, non_blocking=True)
+            T = vol.shape[1] // 3
+            img1 = vol[:, :3]
+            img2 = vol[:, 3:6]
+            mean = torch.as_tensor(mean_bgr, device=vol.device)
+            x = (vol.float() - mean.repeat(T).view(1, -1, 1, 1)) / 255.0
+            out = model(x)
+            flows = out[0] if isinstance(out, tuple) else out
+            mult, cmin, cmax = EVAL_POSTPROC.get(
+                dataset, EVAL_POSTPROC["flying_chairs"])
+            pred = (flows[0][:, :2].float() * flow_scale_finest * mult
+                    ).clamp(cmin, cmax)
+            gt = gt[:, :2]
+            if tuple(pred.shape[-2:]) != tuple(gt.shape[-2:]):
+                pred = ops.resize_bilinear(pred, gt.shape[-2], gt.shape[-1])
+        else:
+            img1 = batch["img1"].to(device, non_blocking=True)
+            img2 = batch["img2"].to(device, non_blocking=True)
+            pred = predict_flow(model, img1, img2, mean_bgr,
+                                flow_scale_finest, dataset,
+                                gt_size=tuple(gt.shape[-2:]))
         total += float(ops.endpoint_error_sum(pred, gt))
         count += gt.shape[0] * gt.shape[-2] * gt.shape[-1]
         if dump_dir is not None and i % dump_every == 0:

@@ -25,8 +25,7 @@ class FusedAdam(Optimizer):
         # cached device chunk table + pinned hyper buffer (hipGraph-safe:
         # the captured step re-reads [lr, bias1, bias2] from pinned host
         # memory at every replay)
-        self._table = None
-        self._table_sig = None
+        self._tables = {}  # group index -> (sig, table, n_chunks)
         self._hyper_pin = None
         self._hyper_dev = None
 
@@ -37,7 +36,7 @@ class FusedAdam(Optimizer):
             with torch.enable_grad():
                 loss = closure()
 
-        for group in self.param_groups:
+        for gi, group in enumerate(self.param_groups):
             params, grads, exp_avgs, exp_avg_sqs = [], [], [], []
             for p in group["params"]:
                 if p.grad is None:
@@ -70,19 +69,22 @@ class FusedAdam(Optimizer):
                 sig = (len(params), params[0].data_ptr(),
                        grads[0].data_ptr(), params[-1].data_ptr(),
                        exp_avgs[0].data_ptr(), exp_avg_sqs[-1].data_ptr())
-                if self._table is None or self._table_sig != sig:
-                    self._table = hip.build_adam_table(
+                cached = self._tables.get(gi)
+                if cached is None or cached[0] != sig:
+                    table = hip.build_adam_table(
                         params, grads, exp_avgs, exp_avg_sqs)
-                    self._table_sig = sig
-                    self._n_chunks = self._table.numel() // 40  # sizeof(ChunkInfo)
-                    self._hyper_pin = torch.zeros(3, pin_memory=True)
-                    self._hyper_dev = torch.zeros(3, device=params[0].device)
+                    cached = (sig, table, table.numel() // 40)  # 40 = sizeof(ChunkInfo)
+                    self._tables[gi] = cached
+                    if self._hyper_pin is None:
+                        self._hyper_pin = torch.zeros(3, pin_memory=True)
+                        self._hyper_dev = torch.zeros(
+                            3, device=params[0].device)
                 self._hyper_pin[0] = group["lr"]
                 self._hyper_pin[1] = bias1
                 self._hyper_pin[2] = bias2
                 self._hyper_dev.copy_(self._hyper_pin, non_blocking=True)
                 hip.fused_adam_table(
-                    self._table, self._n_chunks, self._hyper_dev,
+                    cached[1], cached[2], self._hyper_dev,
                     beta1, beta2, group["eps"], group["weight_decay"],
                 )
             else:

@@ -252,3 +252,26 @@ def test_predict_flow_joint_model():
     pred = predict_flow(m, img1, img2, (104.0, 117.0, 123.0), 10.0,
                         "ucf101", gt_size=(64, 96))
     assert pred.shape == (1, 2, 64, 96)
+
+
+def test_evaluate_aee_volume():
+    from deepof_amd.data import build_dataloader
+    from deepof_amd.engine.evaluator import evaluate_aee
+    from deepof_amd.models import build_model
+    from torch.utils.data import Dataset
+
+    class VolDs(Dataset):
+        def __len__(self):
+            return 2
+
+        def __getitem__(self, i):
+            torch.manual_seed(i)
+            return {"volume": torch.rand(9, 64, 96) * 255,
+                    "flow": torch.randn(4, 64, 96)}
+
+    model, scales, _ = build_model("inception_v3", time_step=3)
+    dl = build_dataloader(VolDs(), 1, shuffle=False, num_workers=0,
+                          drop_last=False)
+    aee = evaluate_aee(model, dl, (70.0, 83.0, 92.0), scales[0], "cpu",
+                       "sintel")
+    assert aee > 0
