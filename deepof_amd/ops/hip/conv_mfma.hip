@@ -492,6 +492,7 @@ void conv_fwd_mfma256_kernel(const bf16* __restrict__ x,
 
   const int frag_col = (lane >> 4) * 8;  // k offset within the 32-col half
 
+  bf16x8 bfrag[4];
   for (int t = 0; t < n_stages; ++t) {
     const int buf = t & 1;
 #pragma unroll
@@ -510,8 +511,10 @@ void conv_fwd_mfma256_kernel(const bf16* __restrict__ x,
       }
 
       // ds-read this phase's fragments (plain loads: hipcc emits the
-      // fine-grained lgkmcnt waits before the MFMAs itself)
-      bf16x8 afrag[4], bfrag[4];
+      // fine-grained lgkmcnt waits before the MFMAs itself).
+      // B fragments are shared by the two m-quadrant phases of a
+      // k-half: read only when the quadrant is 0.
+      bf16x8 afrag[4];
       const bf16* a_base = LDS256_A(buf, kh);
       const bf16* b_base = LDS256_B(buf, kh);
 #pragma unroll
@@ -520,11 +523,14 @@ void conv_fwd_mfma256_kernel(const bf16* __restrict__ x,
         const int col = frag_col ^ (swz256f<SWZV>(row) << 3);
         afrag[mf] = *reinterpret_cast<const bf16x8*>(a_base + row * 32 + col);
       }
+      if (mq == 0) {
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf) {
-        const int row = wn + nf * 16 + (lane & 15);
-        const int col = frag_col ^ (swz256f<SWZV>(row) << 3);
-        bfrag[nf] = *reinterpret_cast<const bf16x8*>(b_base + row * 32 + col);
+        for (int nf = 0; nf < 4; ++nf) {
+          const int row = wn + nf * 16 + (lane & 15);
+          const int col = frag_col ^ (swz256f<SWZV>(row) << 3);
+          bfrag[nf] = *reinterpret_cast<const bf16x8*>(
+              b_base + row * 32 + col);
+        }
       }
 
       // issue next tile's half for THIS phase slot (4 phases ahead)
