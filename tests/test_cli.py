@@ -41,3 +41,25 @@ def test_cli_train_eval_infer(tmp_path):
               "num_workers=0", "device=cpu", f"log_dir={tmp_path}"], repo)
     assert r.returncode == 0, r.stderr[-2000:]
     assert "AEE:" in r.stdout
+
+
+def test_bench_contract_single_process():
+    """bench.py (the driver contract) runs on CPU and prints ONE valid
+    JSON line with the required fields."""
+    import json
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "1", "--warmup", "0",
+         "--batch", "1", "--height", "64", "--width", "64",
+         "--dtype", "fp32", "--no-channels-last", "--no-graphs"],
+        capture_output=True, text=True, cwd=repo, timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    rec = json.loads(line)
+    for field in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                  "ms_per_step", "higher_is_better", "scaling",
+                  "vs_baseline", "dtype", "data", "config"):
+        assert field in rec, field
+    assert rec["n_gpus"] == 1 and rec["scaling"] == "weak"
+    assert rec["config"]["parallelism"] == "dp1"
