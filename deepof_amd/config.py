@@ -64,7 +64,7 @@ class Config:
     # runtime
     precision: str = "bf16"              # bf16|fp32
     device: str = "cuda"
-    channels_last: bool = False
+    channels_last: bool = True   # NHWC: MIOpen NCHW-bf16 falls back to naive kernels
     save_interval_epochs: int = 5
     log_interval: int = 50
     eval_interval_epochs: int = 1
