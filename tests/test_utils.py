@@ -49,3 +49,41 @@ def test_flow_to_color():
     # zero flow -> near-white (saturation ~0 at radius 0)
     white = flow_to_color(np.zeros((4, 4, 2), dtype=np.float32))
     assert (white > 200).all()
+
+
+def test_flo_error_paths(tmp_path):
+    import pytest
+
+    p = tmp_path / "bad.flo"
+    p.write_bytes(b"\x00" * 12)
+    with pytest.raises(ValueError, match="magic"):
+        read_flo(p)
+    # truncated payload
+    import struct
+
+    p2 = tmp_path / "trunc.flo"
+    p2.write_bytes(struct.pack("<fii", 202021.25, 4, 4) + b"\x00" * 8)
+    with pytest.raises(ValueError, match="truncated"):
+        read_flo(p2)
+    with pytest.raises(ValueError, match="H, W, 2"):
+        write_flo(tmp_path / "x.flo", np.zeros((4, 4, 3), dtype=np.float32))
+
+
+def test_config_roundtrip(tmp_path):
+    import yaml
+
+    from deepof_amd.config import Config
+
+    cfg = Config(dataset="sintel", batch_size=7, lambda_smooth=0.0,
+                 image_size=(256, 512))
+    p = tmp_path / "c.yaml"
+    p.write_text(yaml.safe_dump(cfg.to_dict()))
+    back = Config.from_yaml(str(p))
+    assert back.to_dict() == cfg.to_dict()
+    # overrides + unknown-key rejection
+    over = back.apply_overrides(["lr=0.001", "model=flownetc"])
+    assert over.lr == 0.001 and over.model == "flownetc"
+    import pytest
+
+    with pytest.raises(ValueError, match="unknown config key"):
+        back.apply_overrides(["nonsense=1"])
