@@ -345,14 +345,24 @@ class Trainer:
                     return
             self.epoch += 1
             if (self.rank == 0 and val_ds is not None
-                    and self.epoch % cfg.eval_interval_epochs == 0
-                    and hasattr(val_ds[0], "keys") and "flow" in val_ds[0]):
-                aee = evaluate_aee(
-                    self.raw_model, val_loader, self.mean_bgr,
-                    self.flow_scales[0], self.device, cfg.dataset,
-                )
-                self.log_metrics({"epoch": self.epoch, "aee": aee})
-                print(f"[deepof] ***Test: epoch {self.epoch} AEE {aee:.4f}")
+                    and self.epoch % cfg.eval_interval_epochs == 0):
+                sample = val_ds[0]
+                if "flow" in sample:
+                    aee = evaluate_aee(
+                        self.raw_model, val_loader, self.mean_bgr,
+                        self.flow_scales[0], self.device, cfg.dataset,
+                    )
+                    self.log_metrics({"epoch": self.epoch, "aee": aee})
+                    print(f"[deepof] ***Test: epoch {self.epoch} "
+                          f"AEE {aee:.4f}")
+                if "label" in sample and cfg.action_classes > 0:
+                    from .evaluator import evaluate_accuracy
+
+                    acc = evaluate_accuracy(self.raw_model, val_loader,
+                                            self.mean_bgr, self.device)
+                    self.log_metrics({"epoch": self.epoch, "accuracy": acc})
+                    print(f"[deepof] ***Test: epoch {self.epoch} "
+                          f"accuracy {acc:.4f}")
             if self.epoch % cfg.save_interval_epochs == 0:
                 self.save_checkpoint()
         self.save_checkpoint()
