@@ -39,6 +39,8 @@ def main():
                     help="capture the train step in a hipGraph (no gain at\n"
                          "94%% GPU busy; kept as an option)")
     ap.add_argument("--no-graphs", dest="graphs", action="store_false")
+    ap.add_argument("--bucket-mb", type=float, default=25.0,
+                    help="DDP gradient bucket size (MB) for sweep runs")
     args = ap.parse_args()
 
     import torch.distributed as dist
@@ -52,6 +54,12 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    # --gpus must match the actual launch (torchrun --nproc-per-node N):
+    # a silent mismatch would report the wrong whole-job aggregate
+    if world != args.gpus:
+        raise SystemExit(
+            f"bench.py --gpus {args.gpus} but WORLD_SIZE={world}; launch "
+            f"with torch.distributed.run --nproc-per-node {args.gpus}")
     # CPU/gloo fallback exists so the FULL distributed path (DDP wrapper,
     # barriers, reduction, JSON contract) is testable without a GPU
     use_cuda = torch.cuda.is_available()
@@ -72,7 +80,7 @@ def main():
     mean = DATASET_MEANS["flying_chairs"]
     loss_fn = MultiScaleUnsupLoss(flow_scales, weights, mean)
     if world > 1:
-        model = BucketedDataParallel(model)
+        model = BucketedDataParallel(model, bucket_cap_mb=args.bucket_mb)
     opt = FusedAdam(model.parameters(), lr=1.6e-5)
 
     # synthetic FlyingChairs-shaped data, resident on device; a few
@@ -165,15 +173,37 @@ def main():
         dist.barrier()
     if use_cuda:
         torch.cuda.synchronize()
+    # per-step boundaries: CUDA events (async, ~µs each) on GPU, wall
+    # clock on the CPU fallback -> p50/p90 distribution in the JSON
+    if use_cuda:
+        marks = [torch.cuda.Event(enable_timing=True)
+                 for _ in range(args.steps + 1)]
     t0 = time.perf_counter()
     last = None
+    cpu_marks = [t0]
     for i in range(args.steps):
+        if use_cuda:
+            marks[i].record()
         last = timed_step(i)
+        if not use_cuda:
+            cpu_marks.append(time.perf_counter())
     if use_cuda:
+        marks[args.steps].record()
         torch.cuda.synchronize()
     if world > 1:
         dist.barrier()
     t1 = time.perf_counter()
+
+    if use_cuda:
+        step_ms = [marks[i].elapsed_time(marks[i + 1])
+                   for i in range(args.steps)]
+    else:
+        step_ms = [(cpu_marks[i + 1] - cpu_marks[i]) * 1000.0
+                   for i in range(args.steps)]
+    step_ms.sort()
+
+    def pct(p):
+        return step_ms[min(len(step_ms) - 1, int(p * len(step_ms)))]
 
     elapsed = t1 - t0
     if world > 1:
@@ -192,6 +222,10 @@ def main():
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": ms_per_step,
+            "ms_p50": pct(0.50),
+            "ms_p90": pct(0.90),
+            "ms_min": step_ms[0],
+            "ms_max": step_ms[-1],
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,

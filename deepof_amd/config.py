@@ -77,6 +77,16 @@ class Config:
     action_classes: int = 0              # >0 enables the action head
     action_weight: float = 1.0
 
+    def __post_init__(self):
+        if self.augment and self.guided:
+            # the geometric augmentation transforms the frames the
+            # network sees, but batch["flow"] stays in the original
+            # frame geometry -> guided supervision would be wrong
+            raise ValueError(
+                "augment=True with guided=True is unsupported: the GT "
+                "flow is not transformed through the geometric "
+                "augmentation (disable one of them)")
+
     def to_dict(self):
         return dataclasses.asdict(self)
 

@@ -60,7 +60,15 @@ class CudaPrefetcher:
     def __next__(self):
         if self._next is None:
             raise StopIteration
-        torch.cuda.current_stream(self.device).wait_stream(self.stream)
+        cur = torch.cuda.current_stream(self.device)
+        cur.wait_stream(self.stream)
         batch = self._next
+        # tensors were allocated on the side stream but are consumed on
+        # the compute stream: tell the caching allocator, or it may hand
+        # the memory to a later side-stream H2D copy while compute-stream
+        # kernels still read it (standard apex-prefetcher guard)
+        for v in batch.values():
+            if isinstance(v, torch.Tensor):
+                v.record_stream(cur)
         self._preload()
         return batch

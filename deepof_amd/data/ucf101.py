@@ -49,7 +49,14 @@ class UCF101Dataset(Dataset):
                     self.clips.append((clip_dir, self.class_to_idx[c]))
         if not self.clips:
             raise RuntimeError(f"no {split} clips under {frames_root}")
-        self.rng = np.random.default_rng(seed)
+        self.seed = seed
+        # per-clip draw counters: the frame pick is derived from
+        # (seed, idx, visit-count) so repeated visits to one clip
+        # (across epochs) decorrelate, and DataLoader workers — which
+        # each hold a copy of this dataset but serve disjoint indices —
+        # never replay each other's sequences (one shared default_rng
+        # copied into every persistent worker did exactly that).
+        self._visits = np.zeros(len(self.clips), dtype=np.int64)
 
     def __len__(self):
         return len(self.clips)
@@ -62,7 +69,9 @@ class UCF101Dataset(Dataset):
         )
         if len(frames) < 2:
             raise RuntimeError(f"clip {clip_dir} has <2 frames")
-        i = int(self.rng.integers(0, len(frames) - 1))
+        rng = np.random.default_rng((self.seed, idx, int(self._visits[idx])))
+        self._visits[idx] += 1
+        i = int(rng.integers(0, len(frames) - 1))
         img1 = to_chw(load_image(os.path.join(clip_dir, frames[i]),
                                  self.image_size))
         img2 = to_chw(load_image(os.path.join(clip_dir, frames[i + 1]),

@@ -63,12 +63,15 @@ class FusedAdam(Optimizer):
                 from ..ops.functional import require_hip
 
                 hip = require_hip()
-                # moment pointers are part of the signature:
-                # load_state_dict() replaces exp_avg/exp_avg_sq tensors,
-                # which must invalidate the cached device chunk table
+                # EVERY grad pointer is part of the signature: with
+                # zero_grad(set_to_none=True) grads are reallocated each
+                # backward and any one of them can move independently.
+                # Moment pointers too: load_state_dict() replaces the
+                # exp_avg/exp_avg_sq tensors.
                 sig = (len(params), params[0].data_ptr(),
-                       grads[0].data_ptr(), params[-1].data_ptr(),
-                       exp_avgs[0].data_ptr(), exp_avg_sqs[-1].data_ptr())
+                       params[-1].data_ptr(),
+                       exp_avgs[0].data_ptr(), exp_avg_sqs[-1].data_ptr(),
+                       tuple(g.data_ptr() for g in grads))
                 cached = self._tables.get(gi)
                 if cached is None or cached[0] != sig:
                     table = hip.build_adam_table(

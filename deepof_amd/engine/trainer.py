@@ -171,7 +171,11 @@ class Trainer:
         if isinstance(self.model, BucketedDataParallel):
             self.model.zero_grad_buckets()
         else:
-            self.optimizer.zero_grad(set_to_none=True)
+            # keep grads allocated on GPU so the fused-Adam chunk table's
+            # cached device pointers stay valid (set_to_none=True would
+            # reallocate every grad each backward -> table rebuild/step)
+            self.optimizer.zero_grad(
+                set_to_none=self.device.type != "cuda")
 
     # -- the step ---------------------------------------------------------
     def train_step(self, batch) -> dict:
@@ -314,7 +318,12 @@ class Trainer:
                 except FloatingPointError as e:
                     # divergence guard (reference asserts and dies,
                     # flyingChairsTrain.py:203); here: restart from the
-                    # latest checkpoint, bounded retries
+                    # latest checkpoint, bounded retries.  Under DDP a
+                    # per-rank restart would desynchronize the bucket
+                    # all-reduces (other ranks keep training), so
+                    # distributed runs fail fast instead.
+                    if self.world > 1:
+                        raise
                     nan_restarts += 1
                     if (nan_restarts > cfg.nan_restart_limit
                             or not self.try_resume()):
@@ -323,7 +332,10 @@ class Trainer:
                     print(f"[deepof] {e}; restarted from checkpoint "
                           f"({nan_restarts}/{cfg.nan_restart_limit})")
                     break
-                n_imgs += batch["img1"].shape[0] * self.world
+                # batch size from whichever key this mode carries
+                # ("img1" for pairs, "volume" for Sintel multi-frame)
+                first = batch.get("img1", batch.get("volume"))
+                n_imgs += first.shape[0] * self.world
                 steps_done += 1
                 if self.global_step % cfg.log_interval == 0:
                     dt = time.time() - t0
