@@ -35,13 +35,14 @@ def conv(cin: int, cout: int, k: int = 3, stride: int = 1,
 
 
 def deconv(cin: int, cout: int, act: str | None = "elu") -> nn.Module:
-    """4x4 stride-2 transposed conv: exact 2x upsample (out = 2*in)."""
-    layers: list[nn.Module] = [
-        nn.ConvTranspose2d(cin, cout, 4, stride=2, padding=1, bias=True)
-    ]
-    if act is not None:
-        layers.append(make_act(act))
-    return nn.Sequential(*layers) if len(layers) > 1 else layers[0]
+    """4x4 stride-2 transposed conv: exact 2x upsample (out = 2*in).
+
+    On GPU the sub-pixel MFMA kernel runs it as 4 parity stride-1 convs
+    with fused bias+act (ops/deconv.py); params live in the wrapped
+    nn.ConvTranspose2d."""
+    from ..ops.deconv import FusedDeconvAct
+
+    return FusedDeconvAct(cin, cout, act)
 
 
 def bilinear_deconv_weight(cin: int, cout: int, k: int = 4) -> torch.Tensor:
@@ -118,20 +119,30 @@ class FlowDecoder(nn.Module):
             # sintelWrapFlow.py:413-417 with scale=1)
             return deconv(cin, cout, act=a) if factor == 2 else conv(cin, cout, 3, 1, a)
 
+        # concat channels padded to 64-multiples so the decoder's convs,
+        # flow heads and upconvs are MFMA-kernel-eligible (1026/770/386/
+        # 194/98 raw channels would force the MIOpen fallback everywhere)
+        self._concat_pad: list[int] = [0] * self.num_scales
         concat_ch = feat_channels[0]
         for i in range(self.num_scales):
             self.flow_heads.append(conv(concat_ch, flow_channels, 3, act=None))
             if i < self.num_scales - 1:
                 self.upconvs.append(up(concat_ch, up_channels[i], up_factors[i], act))
                 uf = up(flow_channels, flow_channels, up_factors[i], None)
+                from ..ops.deconv import FusedDeconvAct
+
                 if isinstance(uf, nn.ConvTranspose2d):
                     uf._bilinear_init = True
+                elif isinstance(uf, FusedDeconvAct):
+                    uf.deconv._bilinear_init = True
                 elif isinstance(uf, nn.Sequential):
                     uf[0]._bilinear_init = True
                 else:  # FusedConvAct (stride-1 refinement stage)
                     uf.conv._bilinear_init = True
                 self.upflows.append(uf)
-                concat_ch = feat_channels[i + 1] + up_channels[i] + flow_channels
+                raw = feat_channels[i + 1] + up_channels[i] + flow_channels
+                concat_ch = (raw + 63) // 64 * 64
+                self._concat_pad[i + 1] = concat_ch - raw
 
     def forward(self, features: list[torch.Tensor]) -> list[torch.Tensor]:
         assert len(features) == self.num_scales
@@ -146,6 +157,11 @@ class FlowDecoder(nn.Module):
                 up_flow = self.upflows[i](pr)
                 # a 2x deconv of ceil(s/2) overshoots odd skips by 1 px
                 h, w = skip.shape[-2:]
-                x = torch.cat([skip, up_feat[..., :h, :w],
-                               up_flow[..., :h, :w]], dim=1)
+                parts = [skip, up_feat[..., :h, :w], up_flow[..., :h, :w]]
+                if self._concat_pad[i + 1]:
+                    parts.append(skip.new_zeros(
+                        skip.shape[0], self._concat_pad[i + 1], h, w))
+                x = torch.cat(parts, dim=1)
+                if x.device.type == "cuda":
+                    x = x.contiguous(memory_format=torch.channels_last)
         return flows

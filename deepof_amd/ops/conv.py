@@ -72,6 +72,15 @@ class _FusedConvFn(torch.autograd.Function):
                   .contiguous(memory_format=torch.channels_last))
             gx = require_hip().conv2d_fwd(gy, wt, torch.Tensor(), 1, pad, 0)
             need_gx = False
+        # stride-2 backward-data: zero-insertion-free sub-pixel
+        # decomposition (4 parity launches of the strided-out MFMA
+        # kernel), adopted per shape when it measures faster than MIOpen
+        if need_gx and stride == 2:
+            from .deconv import bwd_data_dispatch
+
+            gx = bwd_data_dispatch(gy, x.shape, w, stride, pad)
+            if gx is not None:
+                need_gx = False
 
         # weight grad via the MFMA wrw kernel when it measures faster.
         # r01 status: correct (rel err ~3e-3, bf16-class) but the

@@ -61,7 +61,14 @@ __device__ inline int swz(int row, int col) {
 // Grid: (ceil(M/BM) * ceil(K/BN)) blocks, 256 threads.
 // Requires C % CBK == 0 where CBK = min(C, BK).
 // ---------------------------------------------------------------------
-template <int BM, int BN, int BK, int ACT, bool GLDS>
+// STRIDED=true: sub-pixel (transposed-conv) epilogue — the M grid
+// enumerates quarter-resolution positions (ty, tx) and each output
+// lands at (ty*ostride + off_y, tx*ostride + off_x) of a full-size
+// [B, out_cstride, OHf, OWf] buffer at channel offset out_coff.  Four
+// parity launches of this variant ARE a stride-2 deconv forward or a
+// stride-2 conv backward-data (zero-insertion-free): each parity is a
+// stride-1 conv with its own sub-filter/pad (host builds the plan).
+template <int BM, int BN, int BK, int ACT, bool GLDS, bool STRIDED = false>
 __global__ __launch_bounds__(256)
 void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
                           const bf16* __restrict__ w,
@@ -70,7 +77,10 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
                           const bf16* __restrict__ zero_page,
                           int B, int IH, int IW, int C,
                           int K, int R, int S, int OH, int OW,
-                          int stride, int pad, int n_tiles_n) {
+                          int stride, int pad, int n_tiles_n,
+                          int pad_x = 0, int ostride = 1, int off_y = 0,
+                          int off_x = 0, int OHf = 0, int OWf = 0,
+                          int out_cstride = 0, int out_coff = 0) {
   __shared__ bf16 lds_all[2 * (BM + BN) * BK];
   // pointer-array init from addrspace(3) is rejected; index arithmetic
 #define LDS_A(buf) (lds_all + (buf) * BM * BK)
@@ -90,6 +100,7 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
   const int m0 = tile_m * BM;
   const int n0 = tile_n * BN;
   const int M = B * OH * OW;
+  const int padx = STRIDED ? pad_x : pad;      // asymmetric parity pads
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -131,7 +142,7 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
       const int oy = (mm / OW) % OH;
       const int bb = mm / (OW * OH);
       ga_iy[gi] = oy * stride - pad;
-      ga_ix[gi] = ox * stride - pad;
+      ga_ix[gi] = ox * stride - padx;
       ga_base[gi] = (m < M) ? (long)bb * IH * IW : -1;
     }
   }
@@ -205,7 +216,7 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
       bf16x8 v = {};
       if (a_b[p] >= 0) {
         const int iy = a_oy[p] * stride + r - pad;
-        const int ix = a_ox[p] * stride + s - pad;
+        const int ix = a_ox[p] * stride + s - padx;
         if (iy >= 0 && iy < IH && ix >= 0 && ix < IW) {
           const bf16* src = x + (((long)a_b[p] * IH + iy) * IW + ix) * C + c;
           v = *reinterpret_cast<const bf16x8*>(src);
@@ -270,19 +281,43 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
 
   // ---- epilogue: bias + act, write NHWC ----
   // C/D layout (16x16): col = lane&15, row = (lane>>4)*4 + reg
+  if constexpr (!STRIDED) {
 #pragma unroll
-  for (int mi = 0; mi < M_FRAGS; ++mi) {
+    for (int mi = 0; mi < M_FRAGS; ++mi) {
 #pragma unroll
-    for (int ni = 0; ni < N_FRAGS; ++ni) {
-      const int n = n0 + wn + ni * 16 + (lane & 15);
-      if (n >= K) continue;
-      const float bv = bias ? bias[n] : 0.f;
+      for (int ni = 0; ni < N_FRAGS; ++ni) {
+        const int n = n0 + wn + ni * 16 + (lane & 15);
+        if (n >= K) continue;
+        const float bv = bias ? bias[n] : 0.f;
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int m = m0 + wm + mi * 16 + (lane >> 4) * 4 + reg;
+          if (m >= M) continue;
+          const float val = act_fn<ACT>(acc[mi][ni][reg] + bv);
+          out[(long)m * K + n] = (bf16)val;
+        }
+      }
+    }
+  } else {
+#pragma unroll
+    for (int mi = 0; mi < M_FRAGS; ++mi) {
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         const int m = m0 + wm + mi * 16 + (lane >> 4) * 4 + reg;
         if (m >= M) continue;
-        const float val = act_fn<ACT>(acc[mi][ni][reg] + bv);
-        out[(long)m * K + n] = (bf16)val;
+        const int tx = m % OW;
+        const int ty = (m / OW) % OH;
+        const int bb = m / (OW * OH);
+        const long opix = ((long)bb * OHf + ty * ostride + off_y) * OWf +
+                          tx * ostride + off_x;
+#pragma unroll
+        for (int ni = 0; ni < N_FRAGS; ++ni) {
+          const int n = n0 + wn + ni * 16 + (lane & 15);
+          if (n >= K) continue;
+          const float bv = bias ? bias[n] : 0.f;
+          const float val = act_fn<ACT>(acc[mi][ni][reg] + bv);
+          out[opix * out_cstride + out_coff + n] = (bf16)val;
+        }
       }
     }
   }
@@ -361,6 +396,88 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
 #undef LAUNCH_ACT
 #undef LAUNCH
   return out;
+}
+
+// =====================================================================
+// Sub-pixel strided-output conv: one parity of a transposed conv.
+//
+// Computes a stride-1 conv of x with the (small) parity sub-filter w
+// and writes output position (ty, tx) to pixel (ty*ostride + off_y,
+// tx*ostride + off_x) of `out` [B, out_cstride, OHf, OWf]
+// channels_last, at channel offset out_coff.  Four parity calls
+// = a full 2x deconv forward (decoder upconvs,
+// /root/reference/flyingChairsWrapFlow.py:65-66) or a stride-2 conv
+// backward-data (zero-insertion-free: no dilated gy, no wasted MFMA
+// work on zeros).  The host plan (ops/deconv.py) picks per-parity
+// sub-filters, pads and offsets.
+// =====================================================================
+void conv2d_fwd_strided(at::Tensor x, at::Tensor w, at::Tensor bias,
+                        at::Tensor out, long pad_y, long pad_x, long act,
+                        long ostride, long off_y, long off_x,
+                        long out_coff) {
+  TORCH_CHECK(x.is_cuda() && w.is_cuda() && out.is_cuda());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+              w.scalar_type() == at::kBFloat16 &&
+              out.scalar_type() == at::kBFloat16, "bf16 only");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(out.is_contiguous(at::MemoryFormat::ChannelsLast));
+  const int B = x.size(0), C = x.size(1), IH = x.size(2), IW = x.size(3);
+  const int K = w.size(0), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(w.size(1) == C);
+  TORCH_CHECK(C % 8 == 0, "needs C % 8 == 0, got ", C);
+  constexpr int BM = 128, BK = 64;
+  TORCH_CHECK((R * S * C) % BK == 0, "R*S*C must be a multiple of ", BK);
+  TORCH_CHECK(out.size(0) == B);
+  const int OHf = out.size(2), OWf = out.size(3);
+  const int out_cstride = out.size(1);
+  TORCH_CHECK(out_coff + K <= out_cstride);
+  // the M grid enumerates target pixels ty*ostride+off_y < OHf
+  const int MH = (int)((OHf - off_y + ostride - 1) / ostride);
+  const int MW = (int)((OWf - off_x + ostride - 1) / ostride);
+  const long M = (long)B * MH * MW;
+
+  const float* bptr = nullptr;
+  at::Tensor bias_f;
+  if (bias.defined() && bias.numel()) {
+    bias_f = bias.to(at::kFloat).contiguous();
+    bptr = bias_f.data_ptr<float>();
+  }
+  const bf16* xp = reinterpret_cast<const bf16*>(x.data_ptr());
+  const bf16* wp = reinterpret_cast<const bf16*>(w.data_ptr());
+  bf16* op = reinterpret_cast<bf16*>(out.data_ptr());
+
+  const int BN = (K >= 128) ? 128 : 64;
+  const int n_tiles_n = (K + BN - 1) / BN;
+  const long n_blocks = ((M + BM - 1) / BM) * n_tiles_n;
+  const dim3 grid((unsigned)n_blocks), block(256);
+
+  static at::Tensor zero_page;
+  if (!zero_page.defined() || zero_page.device() != x.device())
+    zero_page = at::zeros({64}, x.options());
+  const bf16* zp = reinterpret_cast<const bf16*>(zero_page.data_ptr());
+  const bool glds = (C % BK == 0);
+
+#define SLAUNCH(BN_, ACT_, GLDS_)                                          \
+  hipLaunchKernelGGL(                                                      \
+      (conv_fwd_mfma_kernel<BM, BN_, BK, ACT_, GLDS_, true>), grid, block, \
+      0, deepof_stream(), xp, wp, bptr, op, zp, B, IH, IW, C, K, R, S,     \
+      MH, MW, 1, (int)pad_y, n_tiles_n, (int)pad_x, (int)ostride,          \
+      (int)off_y, (int)off_x, OHf, OWf, out_cstride, (int)out_coff)
+#define SLAUNCH_ACT(BN_, GLDS_)                                            \
+  do {                                                                     \
+    if (act == 1) SLAUNCH(BN_, 1, GLDS_);                                  \
+    else if (act == 2) SLAUNCH(BN_, 2, GLDS_);                             \
+    else if (act == 3) SLAUNCH(BN_, 3, GLDS_);                             \
+    else SLAUNCH(BN_, 0, GLDS_);                                           \
+  } while (0)
+  if (BN == 128) {
+    if (glds) SLAUNCH_ACT(128, true); else SLAUNCH_ACT(128, false);
+  } else {
+    if (glds) SLAUNCH_ACT(64, true); else SLAUNCH_ACT(64, false);
+  }
+#undef SLAUNCH_ACT
+#undef SLAUNCH
 }
 
 // =====================================================================

@@ -24,6 +24,10 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                       long stride, long pad, long act);
 at::Tensor conv2d_fwd256(at::Tensor x, at::Tensor w, at::Tensor bias,
                          long stride, long pad, long act);
+void conv2d_fwd_strided(at::Tensor x, at::Tensor w, at::Tensor bias,
+                        at::Tensor out, long pad_y, long pad_x, long act,
+                        long ostride, long off_y, long off_x,
+                        long out_coff);
 at::Tensor conv2d_wrw(at::Tensor gy, at::Tensor x, long R, long S,
                       long stride, long pad);
 at::Tensor build_adam_table(std::vector<at::Tensor> params,
@@ -60,4 +64,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "deep-pipelined 256x256 MFMA conv (counted vmcnt, raw barriers)");
   m.def("conv2d_wrw", &conv2d_wrw,
         "MFMA weight gradient (transpose-staged, pixel-split atomics)");
+  m.def("conv2d_fwd_strided", &conv2d_fwd_strided,
+        "sub-pixel strided-output conv (deconv fwd / stride-2 bwd-data "
+        "parity launch)");
 }
