@@ -240,6 +240,16 @@ def main():
                 "parallelism": f"dp{world}",
             },
         }))
+    if rank == 0 and os.environ.get("DEEPOF_DUMP_DISPATCH"):
+        from deepof_amd.ops import conv as _c
+        from deepof_amd.ops import deconv as _d
+
+        for name, cache in (("conv", _c._dispatch_cache),
+                            ("wrw", _c._wrw_cache),
+                            ("deconv", _d._deconv_cache),
+                            ("bwd2", _d._bwd_cache)):
+            for k, v in cache.items():
+                print(f"# dispatch {name}: {k} -> {v}", file=sys.stderr)
     if world > 1:
         dist.destroy_process_group()
 

@@ -292,14 +292,16 @@ def test_act_grad_matches_autograd():
     for code, fn in ((1, torch.nn.functional.elu),
                      (2, lambda t: torch.nn.functional.leaky_relu(t, 0.1)),
                      (3, torch.nn.functional.relu)):
-        pre = torch.randn(2, 16, 8, 10, device=DEV).requires_grad_(True)
+        pre = torch.randn(2, 16, 8, 10, device=DEV,
+                          dtype=torch.bfloat16).requires_grad_(True)
         y = fn(pre)
         g = torch.randn_like(y)
         (want,) = torch.autograd.grad(y, pre, g)
         got = _hip().act_grad(
             g.contiguous(memory_format=torch.channels_last),
             y.detach().contiguous(memory_format=torch.channels_last), code)
-        torch.testing.assert_close(got.float(), want, rtol=1e-4, atol=1e-4)
+        torch.testing.assert_close(got.float(), want.float(),
+                                   rtol=2e-2, atol=2e-2)
 
 
 @pytest.mark.parametrize("swz", [0, 1, 2])
