@@ -48,6 +48,30 @@ class _FusedConvFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, gy):
+        return _conv_backward_impl(ctx, gy) + (None,)
+
+
+class _MiopenConvFn(torch.autograd.Function):
+    """MIOpen forward (it measured faster for this shape), but OUR
+    backward dispatch: fused act-grad + MFMA backward-data kernels."""
+
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, pad, act_code):
+        y = torch.ops.aten.convolution(
+            x, w, bias, [stride, stride], [pad, pad], [1, 1], False,
+            [0, 0], 1)
+        y = _act(y, {v: k for k, v in _ACT_CODE.items()}[act_code])
+        y = y.contiguous(memory_format=torch.channels_last)
+        ctx.save_for_backward(x, w, y)
+        ctx.meta = (stride, pad, act_code, bias is not None)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        return _conv_backward_impl(ctx, gy)
+
+
+def _conv_backward_impl(ctx, gy):
         x, w, y = ctx.saved_tensors
         stride, pad, act_code, has_bias = ctx.meta
         gy = gy.contiguous(memory_format=torch.channels_last)
@@ -106,7 +130,7 @@ class _FusedConvFn(torch.autograd.Function):
             gx = gx2
         if gw is None:
             gw = gw2
-        return gx, gw, gb, None, None, None, None
+        return gx, gw, gb, None, None, None
 
 
 class FusedConvAct(nn.Module):
@@ -143,6 +167,20 @@ class FusedConvAct(nn.Module):
         return _FusedConvFn.apply(x, w, self.conv.bias, self.stride,
                                   self.k // 2, self.act_code, variant)
 
+    def _miopen_fn(self, x):
+        """MIOpen fwd with OUR backward (fused act-grad + sub-pixel
+        stride-2 backward-data dispatch)."""
+        w = self.conv.weight
+        if torch.is_autocast_enabled():
+            x = x.to(torch.bfloat16)
+            w = w.to(torch.bfloat16)
+        if x.dtype != torch.bfloat16 or w.dtype != torch.bfloat16:
+            return None
+        w = w.contiguous(memory_format=torch.channels_last)
+        x = x.contiguous(memory_format=torch.channels_last)
+        return _MiopenConvFn.apply(x, w, self.conv.bias, self.stride,
+                                   self.k // 2, self.act_code)
+
     def forward(self, x):
         if not self._hip_eligible(x):
             return self._miopen(x)
@@ -153,6 +191,13 @@ class FusedConvAct(nn.Module):
             choice = self._autotune(x, key)
         if choice.startswith("hip"):
             y = self._hip(x, choice)
+            if y is not None:
+                return y
+        # MIOpen won the forward; still claim the backward for stride-2
+        # layers where the sub-pixel bwd-data kernel is eligible
+        if (self.stride == 2 and self.conv.out_channels % 64 == 0
+                and torch.is_grad_enabled()):
+            y = self._miopen_fn(x)
             if y is not None:
                 return y
         return self._miopen(x)

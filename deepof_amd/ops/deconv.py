@@ -62,29 +62,69 @@ def subpixel_eligible(red_ch: int, out_ch: int) -> bool:
     return red_ch % 64 == 0 and out_ch >= 8
 
 
-def conv_transpose2d_subpixel(x: torch.Tensor, w_nm: torch.Tensor,
+_tab_cache: dict[tuple, tuple] = {}
+
+
+def _plan_tabs(R: int, S: int, pad: int, N: int, M: int, device):
+    """Device int32 tables for the single-launch path: pack rows
+    {off, nry, nrx, ty0..3, tx0..3, 0} and conv rows
+    {woff, R', S', pad_y, pad_x, off_y, off_x, 0} per parity."""
+    key = (R, S, pad, N, M, str(device))
+    cached = _tab_cache.get(key)
+    if cached is not None:
+        return cached
+    pack_rows, conv_rows = [], []
+    off = 0
+    for uy, taps_y, pad_y in _axis_plan(R, pad):
+        for ux, taps_x, pad_x in _axis_plan(S, pad):
+            nry, nrx = len(taps_y), len(taps_x)
+            assert 1 <= nry <= 4 and 1 <= nrx <= 4
+            pack_rows.append([off, nry, nrx]
+                             + taps_y + [0] * (4 - nry)
+                             + taps_x + [0] * (4 - nrx) + [0])
+            conv_rows.append([off, nry, nrx, pad_y, pad_x, uy, ux, 0])
+            off += N * nry * nrx * M
+    pack_tab = torch.tensor(pack_rows, dtype=torch.int32, device=device)
+    conv_tab = torch.tensor(conv_rows, dtype=torch.int32, device=device)
+    _tab_cache[key] = (pack_tab, conv_tab)
+    return pack_tab, conv_tab
+
+
+def conv_transpose2d_subpixel(x: torch.Tensor, w: torch.Tensor,
                               bias, pad: int, out_h: int, out_w: int,
                               act: int = 0,
                               out: torch.Tensor | None = None,
                               out_coff: int = 0) -> torch.Tensor:
-    """y[b,n,oy,ox] = act(sum_{m,r,s} x[b,m,t,u] w_nm[n,m,r,s] + bias[n])
+    """y[b,n,oy,ox] = act(sum_{m,r,s} x[b,m,t,u] w[m,n,r,s] + bias[n])
     with oy = 2*t - pad + r (transposed-conv scatter semantics).
 
-    x: [B, M, H, W] channels_last bf16; w_nm: [N, M, R, S] bf16;
-    out: optional pre-allocated [B, >=out_coff+N, out_h, out_w]
-    channels_last bf16 buffer (parities cover every pixel, no init
-    needed when writing the full channel range).
+    x: [B, M, H, W] channels_last bf16; w: [M, N, R, S] bf16 — the
+    conv2d weight itself for backward-data, the ConvTranspose2d weight
+    for a deconv forward.  out: optional pre-allocated
+    [B, >=out_coff+N, out_h, out_w] channels_last bf16 buffer (parities
+    cover every pixel, no init needed for the full channel range).
     """
     from .functional import require_hip
 
     hip = require_hip()
-    N, M, R, S = w_nm.shape
+    M, N, R, S = w.shape
     B = x.shape[0]
     if out is None:
         out = torch.empty(
             (B, N, out_h, out_w), device=x.device, dtype=torch.bfloat16
         ).contiguous(memory_format=torch.channels_last)
     b = bias if bias is not None else _empty().to(x.device)
+    if M % 64 == 0:
+        # fast path: 1 pack launch + 1 all-parity conv launch
+        wcl = w.contiguous(memory_format=torch.channels_last)
+        pack_tab, conv_tab = _plan_tabs(R, S, pad, N, M, x.device)
+        wp = hip.subpixel_pack(wcl, pack_tab, R, S)
+        hip.conv2d_fwd_subpixel4(x, wp, b, out, conv_tab, N, act, 2,
+                                 out_coff)
+        return out
+    # generic path: 4 separate parity launches with torch-gathered
+    # sub-filters (odd channel counts)
+    w_nm = w.transpose(0, 1)
     for uy, taps_y, pad_y in _axis_plan(R, pad):
         for ux, taps_x, pad_x in _axis_plan(S, pad):
             if not taps_y or not taps_x or uy >= out_h or ux >= out_w:
@@ -103,16 +143,14 @@ def deconv2d_fwd(x: torch.Tensor, w_ct: torch.Tensor, bias,
     with fused bias + activation.  w_ct: [C_in, C_out, 4, 4]."""
     H, W = x.shape[-2:]
     return conv_transpose2d_subpixel(
-        x, w_ct.transpose(0, 1), bias, 1, 2 * H, 2 * W, act,
-        out=out, out_coff=out_coff)
+        x, w_ct, bias, 1, 2 * H, 2 * W, act, out=out, out_coff=out_coff)
 
 
 def conv2d_bwd_data_subpixel(gy: torch.Tensor, w: torch.Tensor,
                              pad: int, ih: int, iw: int) -> torch.Tensor:
     """dx of a stride-2 conv: dx[b,c,iy,ix] = sum gy[b,k,oy,ox] w[k,c,r,s]
     with iy = 2*oy - pad + r.  w: [K, C, R, S] bf16."""
-    return conv_transpose2d_subpixel(gy, w.transpose(0, 1), None, pad,
-                                     ih, iw, 0)
+    return conv_transpose2d_subpixel(gy, w, None, pad, ih, iw, 0)
 
 
 # -- autograd + module wrapper for decoder upconvs ----------------------

@@ -68,7 +68,14 @@ __device__ inline int swz(int row, int col) {
 // parity launches of this variant ARE a stride-2 deconv forward or a
 // stride-2 conv backward-data (zero-insertion-free): each parity is a
 // stride-1 conv with its own sub-filter/pad (host builds the plan).
-template <int BM, int BN, int BK, int ACT, bool GLDS, bool STRIDED = false>
+// PARITY4=true: all four parities of a sub-pixel transposed conv in ONE
+// launch — blockIdx.y selects the parity, whose geometry {packed-weight
+// offset, R, S, pad_y, pad_x, off_y, off_x} comes from the 8-int rows
+// of `ptab` (device memory, built once per layer plan).  This removes
+// the 4-launch + per-launch weight-gather overhead that dominates the
+// small decoder layers.
+template <int BM, int BN, int BK, int ACT, bool GLDS, bool STRIDED = false,
+          bool PARITY4 = false>
 __global__ __launch_bounds__(256)
 void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
                           const bf16* __restrict__ w,
@@ -80,7 +87,8 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
                           int stride, int pad, int n_tiles_n,
                           int pad_x = 0, int ostride = 1, int off_y = 0,
                           int off_x = 0, int OHf = 0, int OWf = 0,
-                          int out_cstride = 0, int out_coff = 0) {
+                          int out_cstride = 0, int out_coff = 0,
+                          const int* __restrict__ ptab = nullptr) {
   __shared__ bf16 lds_all[2 * (BM + BN) * BK];
   // pointer-array init from addrspace(3) is rejected; index arithmetic
 #define LDS_A(buf) (lds_all + (buf) * BM * BK)
@@ -94,6 +102,18 @@ void conv_fwd_mfma_kernel(const bf16* __restrict__ x,
     const int q = nwg / 8, rr = nwg % 8;
     const int xcd = bid % 8, idx = bid / 8;
     bid = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  if constexpr (PARITY4) {
+    const int* pp = ptab + blockIdx.y * 8;
+    w += pp[0];
+    R = pp[1];
+    S = pp[2];
+    pad = pp[3];
+    pad_x = pp[4];
+    off_y = pp[5];
+    off_x = pp[6];
+    OH = (OHf - off_y + ostride - 1) / ostride;  // parity M grid
+    OW = (OWf - off_x + ostride - 1) / ostride;
   }
   const int tile_m = bid / n_tiles_n;
   const int tile_n = bid % n_tiles_n;
@@ -478,6 +498,132 @@ void conv2d_fwd_strided(at::Tensor x, at::Tensor w, at::Tensor bias,
   }
 #undef SLAUNCH_ACT
 #undef SLAUNCH
+}
+
+// =====================================================================
+// Sub-pixel parity weight pack: gather the 4 parity sub-filters of a
+// transposed conv out of the live weight in ONE launch.
+//   src: channels_last [M, N, R, S]  (memory [M][R][S][N]) — for
+//        bwd-data this is the conv weight [K, C, R, S] itself; for a
+//        deconv the ConvTranspose2d weight [C, K, 4, 4] channels_last.
+//   out: flat; parity p at tab[p*12+0] with layout [N][jy][jx][M]
+//        (= the conv kernel's [Kout, R', S', Cred] B-operand layout).
+//   tab (device int32 [4,12]): {off, nry, nrx, ty0..3, tx0..3, pad}.
+// Reads are wave-coalesced along N; each thread writes one 16-B
+// bf16x8 chunk of the M-contiguous output row.
+// =====================================================================
+namespace {
+
+__global__ __launch_bounds__(256)
+void subpixel_pack_kernel(const bf16* __restrict__ w, bf16* __restrict__ out,
+                          const int* __restrict__ tab, int N, int M,
+                          int R, int S) {
+  const int* t = tab + blockIdx.y * 12;
+  const int off = t[0], nry = t[1], nrx = t[2];
+  const int M8 = M >> 3;
+  const long total = (long)N * M8 * nry * nrx;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const int n = (int)(i % N);
+    long r = i / N;
+    const int m8 = (int)(r % M8);
+    r /= M8;
+    const int jx = (int)(r % nrx);
+    const int jy = (int)(r / nrx);
+    const int ty = t[3 + jy], tx = t[7 + jx];
+    bf16x8 v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      v[j] = w[(((long)(m8 * 8 + j) * R + ty) * S + tx) * N + n];
+    *reinterpret_cast<bf16x8*>(
+        out + off + (((long)n * nry + jy) * nrx + jx) * M + m8 * 8) = v;
+  }
+}
+
+}  // namespace
+
+at::Tensor subpixel_pack(at::Tensor w, at::Tensor tab, long R_, long S_) {
+  TORCH_CHECK(w.is_cuda() && w.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "pack source must be channels_last [M, N, R, S]");
+  TORCH_CHECK(tab.is_cuda() && tab.scalar_type() == at::kInt &&
+              tab.numel() == 48);
+  const int M = w.size(0), N = w.size(1);
+  const int R = (int)R_, S = (int)S_;
+  TORCH_CHECK(w.size(2) == R && w.size(3) == S);
+  TORCH_CHECK(M % 8 == 0, "pack needs M % 8 == 0");
+  auto out = at::empty({(long)M * N * R * S}, w.options());
+  const long per_par = (long)N * (M / 8) * R * S;  // upper bound of work
+  const int blocks = (int)std::min((per_par + 255) / 256, (long)1024);
+  hipLaunchKernelGGL(subpixel_pack_kernel, dim3(blocks, 4), dim3(256), 0,
+                     deepof_stream(),
+                     reinterpret_cast<const bf16*>(w.data_ptr()),
+                     reinterpret_cast<bf16*>(out.data_ptr()),
+                     tab.data_ptr<int>(), N, M, R, S);
+  return out;
+}
+
+// =====================================================================
+// All-parity sub-pixel transposed conv in one launch (PARITY4 variant).
+//   x: [B, C, IH, IW] channels_last bf16 (the transposed conv's input;
+//      gy for backward-data);  wpacked: subpixel_pack output;
+//   ptab (device int32 [4,8]): {woff, R, S, pad_y, pad_x, off_y, off_x}.
+// =====================================================================
+void conv2d_fwd_subpixel4(at::Tensor x, at::Tensor wpacked, at::Tensor bias,
+                          at::Tensor out, at::Tensor ptab, long K_, long act,
+                          long ostride, long out_coff) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(out.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(ptab.is_cuda() && ptab.scalar_type() == at::kInt &&
+              ptab.numel() == 32);
+  const int B = x.size(0), C = x.size(1), IH = x.size(2), IW = x.size(3);
+  const int K = (int)K_;
+  TORCH_CHECK(C % 64 == 0, "subpixel4 needs C % 64 == 0 (red channels)");
+  const int OHf = out.size(2), OWf = out.size(3);
+  const int out_cstride = out.size(1);
+  TORCH_CHECK(out_coff + K <= out_cstride);
+  constexpr int BM = 128;
+  const int MH = (OHf + (int)ostride - 1) / (int)ostride;  // max over parities
+  const int MW = (OWf + (int)ostride - 1) / (int)ostride;
+  const long M = (long)B * MH * MW;
+
+  const float* bptr = nullptr;
+  at::Tensor bias_f;
+  if (bias.defined() && bias.numel()) {
+    bias_f = bias.to(at::kFloat).contiguous();
+    bptr = bias_f.data_ptr<float>();
+  }
+  static at::Tensor zero_page;
+  if (!zero_page.defined() || zero_page.device() != x.device())
+    zero_page = at::zeros({64}, x.options());
+
+  const bf16* xp = reinterpret_cast<const bf16*>(x.data_ptr());
+  const bf16* wp = reinterpret_cast<const bf16*>(wpacked.data_ptr());
+  bf16* op = reinterpret_cast<bf16*>(out.data_ptr());
+  const bf16* zp = reinterpret_cast<const bf16*>(zero_page.data_ptr());
+
+  const int BN = (K >= 128) ? 128 : 64;
+  const int n_tiles_n = (K + BN - 1) / BN;
+  const long n_blocks = ((M + BM - 1) / BM) * n_tiles_n;
+  const dim3 grid((unsigned)n_blocks, 4), block(256);
+
+#define P4LAUNCH(BN_, ACT_)                                                 \
+  hipLaunchKernelGGL(                                                       \
+      (conv_fwd_mfma_kernel<BM, BN_, 64, ACT_, true, true, true>), grid,    \
+      block, 0, deepof_stream(), xp, wp, bptr, op, zp, B, IH, IW, C, K,     \
+      0, 0, MH, MW, 1, 0, n_tiles_n, 0, (int)ostride, 0, 0, OHf, OWf,       \
+      out_cstride, (int)out_coff, ptab.data_ptr<int>())
+#define P4LAUNCH_ACT(BN_)                                                   \
+  do {                                                                      \
+    if (act == 1) P4LAUNCH(BN_, 1);                                         \
+    else if (act == 2) P4LAUNCH(BN_, 2);                                    \
+    else if (act == 3) P4LAUNCH(BN_, 3);                                    \
+    else P4LAUNCH(BN_, 0);                                                  \
+  } while (0)
+  if (BN == 128) P4LAUNCH_ACT(128); else P4LAUNCH_ACT(64);
+#undef P4LAUNCH_ACT
+#undef P4LAUNCH
 }
 
 // =====================================================================

@@ -28,6 +28,10 @@ void conv2d_fwd_strided(at::Tensor x, at::Tensor w, at::Tensor bias,
                         at::Tensor out, long pad_y, long pad_x, long act,
                         long ostride, long off_y, long off_x,
                         long out_coff);
+at::Tensor subpixel_pack(at::Tensor w, at::Tensor tab, long R, long S);
+void conv2d_fwd_subpixel4(at::Tensor x, at::Tensor wpacked, at::Tensor bias,
+                          at::Tensor out, at::Tensor ptab, long K, long act,
+                          long ostride, long out_coff);
 at::Tensor conv2d_wrw(at::Tensor gy, at::Tensor x, long R, long S,
                       long stride, long pad);
 at::Tensor build_adam_table(std::vector<at::Tensor> params,
@@ -67,4 +71,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd_strided", &conv2d_fwd_strided,
         "sub-pixel strided-output conv (deconv fwd / stride-2 bwd-data "
         "parity launch)");
+  m.def("subpixel_pack", &subpixel_pack,
+        "gather all 4 parity sub-filters of a transposed conv");
+  m.def("conv2d_fwd_subpixel4", &conv2d_fwd_subpixel4,
+        "all-parity sub-pixel transposed conv in one launch");
 }
