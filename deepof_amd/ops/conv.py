@@ -195,8 +195,11 @@ class FusedConvAct(nn.Module):
                 return y
         # MIOpen won the forward; still claim the backward for stride-2
         # layers where the sub-pixel bwd-data kernel is eligible
+        import os as _os
+
         if (self.stride == 2 and self.conv.out_channels % 64 == 0
-                and torch.is_grad_enabled()):
+                and torch.is_grad_enabled()
+                and not _os.environ.get("DEEPOF_NO_MIOPENFN")):
             y = self._miopen_fn(x)
             if y is not None:
                 return y

@@ -232,6 +232,10 @@ class FusedDeconvAct(torch.nn.Module):
         return _FusedDeconvFn.apply(x, w, self.deconv.bias, self.act_code)
 
     def forward(self, x):
+        import os
+
+        if os.environ.get("DEEPOF_NO_DECONV"):
+            return self._torch(x)
         if not (x.is_cuda and subpixel_eligible(self.cin, self.cout)
                 and x.is_contiguous(memory_format=torch.channels_last)):
             return self._torch(x)
@@ -279,6 +283,10 @@ def bwd_data_dispatch(gy: torch.Tensor, x_shape, w: torch.Tensor,
                       stride: int, pad: int):
     """Returns dx via the sub-pixel kernel when it measures faster than
     MIOpen's backward-data for this shape; None to fall back."""
+    import os
+
+    if os.environ.get("DEEPOF_NO_BWD2"):
+        return None
     if stride != 2 or not subpixel_eligible(w.shape[0], w.shape[1]):
         return None
     ih, iw = x_shape[-2], x_shape[-1]
