@@ -19,7 +19,16 @@ def main():
     window_ms = float(sys.argv[2]) if len(sys.argv) > 2 else 180.0
     n = int(sys.argv[3]) if len(sys.argv) > 3 else 40
     files = sorted(glob.glob(f"{d}/**/*kernel_trace.csv", recursive=True))
-    assert files, f"no kernel trace under {d}"
+    if not files:
+        # tolerate other rocprofv3 naming: any csv with kernel columns
+        for c in sorted(glob.glob(f"{d}/**/*.csv", recursive=True)):
+            with open(c) as f:
+                head = f.readline()
+            if "Kernel_Name" in head and ("Start_Timestamp" in head
+                                          or "BeginNs" in head):
+                files.append(c)
+    assert files, f"no kernel trace under {d}: " + str(
+        glob.glob(f"{d}/**/*", recursive=True)[:20])
     rows = []
     with open(files[-1]) as f:
         rd = csv.DictReader(f)
