@@ -167,9 +167,9 @@ class FlowDecoder(nn.Module):
                     fmt = (torch.channels_last if skip.is_contiguous(
                         memory_format=torch.channels_last)
                         else torch.contiguous_format)
-                    parts.append(torch.zeros(
+                    parts.append(torch.empty(
                         (skip.shape[0], self._concat_pad[i + 1], h, w),
                         dtype=skip.dtype, device=skip.device,
-                        memory_format=fmt))
+                        memory_format=fmt).zero_())
                 x = torch.cat(parts, dim=1)
         return flows
