@@ -34,6 +34,8 @@ void conv2d_fwd_subpixel4(at::Tensor x, at::Tensor wpacked, at::Tensor bias,
                           long ostride, long out_coff);
 at::Tensor conv2d_wrw(at::Tensor gy, at::Tensor x, long R, long S,
                       long stride, long pad);
+at::Tensor conv2d_wrw2(at::Tensor gy, at::Tensor x, long R, long S,
+                       long stride, long pad);
 at::Tensor build_adam_table(std::vector<at::Tensor> params,
                             std::vector<at::Tensor> grads,
                             std::vector<at::Tensor> exp_avgs,
@@ -71,6 +73,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd_strided", &conv2d_fwd_strided,
         "sub-pixel strided-output conv (deconv fwd / stride-2 bwd-data "
         "parity launch)");
+  m.def("conv2d_wrw2", &conv2d_wrw2,
+        "MFMA weight gradient v2 (natural-layout staging, tr_b16 reads)");
   m.def("subpixel_pack", &subpixel_pack,
         "gather all 4 parity sub-filters of a transposed conv");
   m.def("conv2d_fwd_subpixel4", &conv2d_fwd_subpixel4,

@@ -25,6 +25,7 @@ ext = CUDAExtension(
         os.path.join(SRC, "adam.hip"),
         os.path.join(SRC, "conv_mfma.hip"),
         os.path.join(SRC, "conv_wrw.hip"),
+        os.path.join(SRC, "conv_wrw2.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
