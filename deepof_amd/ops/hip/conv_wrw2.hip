@@ -106,14 +106,21 @@ void conv_wrw2_kernel(const bf16* __restrict__ gy,  // [B,OH,OW,K]
   };
 
   // tr-read one 8-pixel fragment: rows ch0..ch0+15 (one ks subtile),
-  // reduction pixels p0..p0+7.  per-lane element address
-  //   ks*SS + (p0 + 8*(lane>>4))*16 + (lane&15)
-  // yields lane l elem j = [ch = ch0 + (l&15)][pix = p0 + 8*(l>>4)+j]
-  // for j=0..3; the +64-element offset read gives j=4..7.
+  // reduction pixels p0..p0+7.  Measured ds_read_b64_tr_b16 semantics
+  // (tools/tr_probe.hip): each lane supplies an 8-B-aligned address of
+  // a 4-element bf16 run; the result is the cross-lane transpose
+  //   out[l][j] = lds[addr_of_lane(16*(l>>4) + 4*j + ((l&15)>>2))
+  //                   + (l&3)].
+  // With per-lane address (elements)
+  //   ks*SS + (p0 + 8*(l>>4) + ((l>>2)&3))*16 + 4*(l&3)
+  // into the [pix][16ch] image, lane l elem j delivers
+  // A[ch0 + (l&15)][p0 + 8*(l>>4) + j] — exactly the 16x16x32 MFMA
+  // operand layout; the +64-element read supplies j = 4..7.
   auto tr_frag = [&](const bf16* img, int ch0, int p0) -> bf16x8 {
     const int ks = ch0 >> 4;
-    const bf16* base = img + ks * SS<BKP> + (p0 + 8 * (lane >> 4)) * 16 +
-                       (lane & 15);
+    const bf16* base = img + ks * SS<BKP> +
+                       (p0 + 8 * (lane >> 4) + ((lane >> 2) & 3)) * 16 +
+                       4 * (lane & 3);
     bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
         (lds_bf16x4*)base);
     bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
@@ -196,7 +203,10 @@ at::Tensor conv2d_wrw2(at::Tensor gy, at::Tensor x, long R_, long S_,
   auto dw_f32 = at::zeros({K, RSC}, x.options().dtype(at::kFloat));
   const int ktiles = (K + 63) / 64;
   const int ntiles = RSC / 64;
-  constexpr int BKP = 64;
+  static const int BKP = [] {
+    const char* e = getenv("DEEPOF_WRW2_BKP");
+    return e ? atoi(e) : 128;
+  }();
   // pixel split so the grid fills the chip (~2 blocks/CU)
   int split = (int)std::max(1L, 512L / ((long)ktiles * ntiles));
   const int pix_per_slice =
@@ -204,13 +214,16 @@ at::Tensor conv2d_wrw2(at::Tensor gy, at::Tensor x, long R_, long S_,
   split = (int)((M + pix_per_slice - 1) / pix_per_slice);
 
   const dim3 grid(ktiles * ntiles, split), block(256);
-  hipLaunchKernelGGL((conv_wrw2_kernel<BKP>), grid, block, 0,
-                     deepof_stream(),
-                     reinterpret_cast<const bf16*>(gy.data_ptr()),
-                     reinterpret_cast<const bf16*>(x.data_ptr()),
-                     dw_f32.data_ptr<float>(), B, IH, IW, C, K, R, S,
-                     OH, OW, (int)stride, (int)pad, ntiles,
-                     pix_per_slice);
+#define WRW2_LAUNCH(BKP_)                                                  \
+  hipLaunchKernelGGL((conv_wrw2_kernel<BKP_>), grid, block, 0,             \
+                     deepof_stream(),                                      \
+                     reinterpret_cast<const bf16*>(gy.data_ptr()),         \
+                     reinterpret_cast<const bf16*>(x.data_ptr()),          \
+                     dw_f32.data_ptr<float>(), B, IH, IW, C, K, R, S,      \
+                     OH, OW, (int)stride, (int)pad, ntiles,                \
+                     pix_per_slice)
+  if (BKP == 128) WRW2_LAUNCH(128); else WRW2_LAUNCH(64);
+#undef WRW2_LAUNCH
   auto dw = dw_f32.view({K, R, S, C}).permute({0, 3, 1, 2}).to(at::kBFloat16);
   return dw.contiguous(at::MemoryFormat::ChannelsLast);
 }
