@@ -106,16 +106,15 @@ def _conv_backward_impl(ctx, gy):
             if gx is not None:
                 need_gx = False
 
-        # weight grad via the MFMA wrw kernel when it measures faster.
-        # r01 status: correct (rel err ~3e-3, bf16-class) but the
-        # transpose-staged 64x64 tile runs 0.25-0.9x MIOpen's igemm wrw
-        # (tools/bench_wrw.py), so the candidate is opt-in until the
-        # staging uses ds_read_tr16 / a bigger tile (round-2 work).
+        # weight grad via the MFMA wrw2 kernel (natural-layout staging
+        # + ds_read_b64_tr_b16 transpose reads) when it measures faster
+        # than MIOpen for this shape (tools/bench_wrw2.py; it wins on
+        # the small-spatial conv6-class shapes).  DEEPOF_WRW=0 disables.
         import os as _os
 
         need_gw = ctx.needs_input_grad[1]
         gw = None
-        if (need_gw and _os.environ.get("DEEPOF_WRW") == "1"
+        if (need_gw and _os.environ.get("DEEPOF_WRW") != "0"
                 and x.shape[1] % 64 == 0 and w.shape[0] % 8 == 0):
             gw = _maybe_hip_wrw(gy, x, w, stride, pad)
             if gw is not None:
@@ -251,7 +250,7 @@ def _maybe_hip_wrw(gy, x, w, stride, pad):
     R, S = w.shape[2], w.shape[3]
     if choice is None:
         def ours():
-            return hip.conv2d_wrw(gy, x, R, S, stride, pad)
+            return hip.conv2d_wrw2(gy, x, R, S, stride, pad)
 
         def mio():
             return torch.ops.aten.convolution_backward(
@@ -273,5 +272,5 @@ def _maybe_hip_wrw(gy, x, w, stride, pad):
         choice = "hip" if t_h < t_m else "miopen"
         _wrw_cache[key] = choice
     if choice == "hip":
-        return hip.conv2d_wrw(gy, x, R, S, stride, pad)
+        return hip.conv2d_wrw2(gy, x, R, S, stride, pad)
     return None

@@ -32,8 +32,6 @@ at::Tensor subpixel_pack(at::Tensor w, at::Tensor tab, long R, long S);
 void conv2d_fwd_subpixel4(at::Tensor x, at::Tensor wpacked, at::Tensor bias,
                           at::Tensor out, at::Tensor ptab, long K, long act,
                           long ostride, long out_coff);
-at::Tensor conv2d_wrw(at::Tensor gy, at::Tensor x, long R, long S,
-                      long stride, long pad);
 at::Tensor conv2d_wrw2(at::Tensor gy, at::Tensor x, long R, long S,
                        long stride, long pad);
 at::Tensor build_adam_table(std::vector<at::Tensor> params,
@@ -68,8 +66,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA implicit-GEMM conv + bias + act (NHWC bf16)");
   m.def("conv2d_fwd256", &conv2d_fwd256,
         "deep-pipelined 256x256 MFMA conv (counted vmcnt, raw barriers)");
-  m.def("conv2d_wrw", &conv2d_wrw,
-        "MFMA weight gradient (transpose-staged, pixel-split atomics)");
   m.def("conv2d_fwd_strided", &conv2d_fwd_strided,
         "sub-pixel strided-output conv (deconv fwd / stride-2 bwd-data "
         "parity launch)");

@@ -24,7 +24,6 @@ ext = CUDAExtension(
         os.path.join(SRC, "correlation.hip"),
         os.path.join(SRC, "adam.hip"),
         os.path.join(SRC, "conv_mfma.hip"),
-        os.path.join(SRC, "conv_wrw.hip"),
         os.path.join(SRC, "conv_wrw2.hip"),
     ],
     extra_compile_args={

@@ -327,7 +327,7 @@ def test_conv2d_fwd256_swizzle_variants(swz, monkeypatch):
     torch.testing.assert_close(got, want, rtol=0.06, atol=0.06)
 
 
-def test_conv2d_wrw_matches_autograd():
+def test_conv2d_wrw2_matches_autograd():
     torch.manual_seed(0)
     B, C, K, H, W = 2, 64, 16, 16, 20
     x = torch.randn(B, C, H, W, device=DEV)
@@ -335,7 +335,7 @@ def test_conv2d_wrw_matches_autograd():
     y = torch.nn.functional.conv2d(x, w, stride=1, padding=1)
     gy = torch.randn_like(y)
     (gw_want,) = torch.autograd.grad(y, w, gy)
-    gw = _hip().conv2d_wrw(
+    gw = _hip().conv2d_wrw2(
         gy.bfloat16().contiguous(memory_format=torch.channels_last),
         x.detach().bfloat16().contiguous(memory_format=torch.channels_last),
         3, 3, 1, 1)

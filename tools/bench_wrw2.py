@@ -44,9 +44,6 @@ def check(name, B, C, K, IH, IW, R, stride, pad, perf=True):
     def ours():
         return hip.conv2d_wrw2(gyb, xb, R, R, stride, pad)
 
-    def v1():
-        return hip.conv2d_wrw(gyb, xb, R, R, stride, pad)
-
     def mio():
         return torch.ops.aten.convolution_backward(
             gyb, xb, cl(w.detach().bfloat16()), None, [stride, stride],
@@ -64,8 +61,8 @@ def check(name, B, C, K, IH, IW, R, stride, pad, perf=True):
         for r, s in bad.tolist()[:4]:
             print(f"   tap({r},{s}): max err {per_tap[r, s].item():.4f}")
     if perf:
-        t2, t1_, tm = timeit(ours), timeit(v1), timeit(mio)
-        print(f"  v2 {t2:7.3f} ms  v1 {t1_:7.3f} ms  miopen {tm:7.3f} ms"
+        t2, tm = timeit(ours), timeit(mio)
+        print(f"  v2 {t2:7.3f} ms  miopen {tm:7.3f} ms"
               f"  v2 speedup x{tm/t2:.2f}")
 
 
