@@ -26,6 +26,9 @@ def main():
     ap.add_argument("--width", type=int, default=128)
     ap.add_argument("--batch", type=int, default=32)
     ap.add_argument("--lr", type=float, default=1e-4)
+    ap.add_argument("--guided", action="store_true",
+                    help="add proxy-label supervision (BASELINE configs[3] "
+                         "guided + warp-photometric loss)")
     args = ap.parse_args()
 
     import torch
@@ -43,6 +46,7 @@ def main():
         run_name="accuracy_run", lr=args.lr, epochs_per_decay=15,
         max_epochs=args.epochs, log_interval=50, eval_interval_epochs=5,
         save_interval_epochs=10, resume=False, seed=0,
+        guided=args.guided, guided_weight=1.0,
     ))
     tr = Trainer(cfg)
     t0 = time.time()
@@ -80,7 +84,7 @@ def main():
         "minutes": round((time.time() - t0) / 60, 2),
         "config": {"model": "flownets", "image": [args.height, args.width],
                    "batch": args.batch, "lr": args.lr,
-                   "loss": "unsupervised photometric+smoothness",
+                   "loss": ("guided+photometric" if args.guided else "unsupervised photometric+smoothness"),
                    "data": "synthetic (GT flow = photometric minimum)"},
     }
     print(json.dumps(result))
