@@ -90,6 +90,12 @@ def main():
     print(json.dumps(result))
     with open("gpurun_out/accuracy_run/result.json", "w") as f:
         json.dump(result, f, indent=1)
+    # checkpoints are ~0.5 GB (model + Adam moments): drop them so the
+    # gpurun copy-back stays under its 64 MiB limit
+    import glob
+
+    for p in glob.glob("gpurun_out/accuracy_run/ckpt_*.pt"):
+        os.remove(p)
 
 
 if __name__ == "__main__":
