@@ -72,64 +72,64 @@ class _MiopenConvFn(torch.autograd.Function):
 
 
 def _conv_backward_impl(ctx, gy):
-        x, w, y = ctx.saved_tensors
-        stride, pad, act_code, has_bias = ctx.meta
-        gy = gy.contiguous(memory_format=torch.channels_last)
-        if act_code in (1, 2, 3):
-            from .functional import require_hip
+    x, w, y = ctx.saved_tensors
+    stride, pad, act_code, has_bias = ctx.meta
+    gy = gy.contiguous(memory_format=torch.channels_last)
+    if act_code in (1, 2, 3):
+        from .functional import require_hip
 
-            # one fused pass (gy * act_grad(y)) instead of where+mul
-            gy = require_hip().act_grad(gy, y, act_code)
+        # one fused pass (gy * act_grad(y)) instead of where+mul
+        gy = require_hip().act_grad(gy, y, act_code)
 
-        need_gx = ctx.needs_input_grad[0]
-        gx = None
-        k = w.shape[-1]
-        # stride-1 backward-data IS a forward conv with the transposed,
-        # spatially flipped weight (dx = gy * W^T_rot) -> reuse the MFMA
-        # kernel when eligible (K*k*k % 64 == 0, same-size output)
-        if (need_gx and stride == 1 and w.shape[0] % 8 == 0
-                and (w.shape[0] * k * k) % 64 == 0
-                and x.shape[-2:] == gy.shape[-2:]):
-            from .functional import require_hip
+    need_gx = ctx.needs_input_grad[0]
+    gx = None
+    k = w.shape[-1]
+    # stride-1 backward-data IS a forward conv with the transposed,
+    # spatially flipped weight (dx = gy * W^T_rot) -> reuse the MFMA
+    # kernel when eligible (K*k*k % 64 == 0, same-size output)
+    if (need_gx and stride == 1 and w.shape[0] % 8 == 0
+            and (w.shape[0] * k * k) % 64 == 0
+            and x.shape[-2:] == gy.shape[-2:]):
+        from .functional import require_hip
 
-            wt = (w.transpose(0, 1).flip(-1, -2)
-                  .contiguous(memory_format=torch.channels_last))
-            gx = require_hip().conv2d_fwd(gy, wt, torch.Tensor(), 1, pad, 0)
+        wt = (w.transpose(0, 1).flip(-1, -2)
+              .contiguous(memory_format=torch.channels_last))
+        gx = require_hip().conv2d_fwd(gy, wt, torch.Tensor(), 1, pad, 0)
+        need_gx = False
+    # stride-2 backward-data: zero-insertion-free sub-pixel
+    # decomposition (4 parity launches of the strided-out MFMA
+    # kernel), adopted per shape when it measures faster than MIOpen
+    if need_gx and stride == 2:
+        from .deconv import bwd_data_dispatch
+
+        gx = bwd_data_dispatch(gy, x.shape, w, stride, pad)
+        if gx is not None:
             need_gx = False
-        # stride-2 backward-data: zero-insertion-free sub-pixel
-        # decomposition (4 parity launches of the strided-out MFMA
-        # kernel), adopted per shape when it measures faster than MIOpen
-        if need_gx and stride == 2:
-            from .deconv import bwd_data_dispatch
 
-            gx = bwd_data_dispatch(gy, x.shape, w, stride, pad)
-            if gx is not None:
-                need_gx = False
+    # weight grad via the MFMA wrw2 kernel (natural-layout staging
+    # + ds_read_b64_tr_b16 transpose reads) when it measures faster
+    # than MIOpen for this shape (tools/bench_wrw2.py; it wins on
+    # the small-spatial conv6-class shapes).  DEEPOF_WRW=0 disables.
+    import os as _os
 
-        # weight grad via the MFMA wrw2 kernel (natural-layout staging
-        # + ds_read_b64_tr_b16 transpose reads) when it measures faster
-        # than MIOpen for this shape (tools/bench_wrw2.py; it wins on
-        # the small-spatial conv6-class shapes).  DEEPOF_WRW=0 disables.
-        import os as _os
+    need_gw = ctx.needs_input_grad[1]
+    gw = None
+    if (need_gw and _os.environ.get("DEEPOF_WRW") != "0"
+            and x.shape[1] % 64 == 0 and w.shape[0] % 8 == 0):
+        gw = _maybe_hip_wrw(gy, x, w, stride, pad)
+        if gw is not None:
+            need_gw = False
 
-        need_gw = ctx.needs_input_grad[1]
-        gw = None
-        if (need_gw and _os.environ.get("DEEPOF_WRW") != "0"
-                and x.shape[1] % 64 == 0 and w.shape[0] % 8 == 0):
-            gw = _maybe_hip_wrw(gy, x, w, stride, pad)
-            if gw is not None:
-                need_gw = False
-
-        gx2, gw2, gb = torch.ops.aten.convolution_backward(
-            gy, x, w, [w.shape[0]] if has_bias else None,
-            [stride, stride], [pad, pad], [1, 1], False, [0, 0], 1,
-            [need_gx, need_gw, has_bias and ctx.needs_input_grad[2]],
-        )
-        if gx is None:
-            gx = gx2
-        if gw is None:
-            gw = gw2
-        return gx, gw, gb, None, None, None
+    gx2, gw2, gb = torch.ops.aten.convolution_backward(
+        gy, x, w, [w.shape[0]] if has_bias else None,
+        [stride, stride], [pad, pad], [1, 1], False, [0, 0], 1,
+        [need_gx, need_gw, has_bias and ctx.needs_input_grad[2]],
+    )
+    if gx is None:
+        gx = gx2
+    if gw is None:
+        gw = gw2
+    return gx, gw, gb, None, None, None
 
 
 class FusedConvAct(nn.Module):
