@@ -118,5 +118,17 @@ class Config:
             k, v = ov.split("=", 1)
             if k not in data:
                 raise ValueError(f"unknown config key {k!r}")
-            data[k] = yaml.safe_load(v)
+            parsed = yaml.safe_load(v)
+            # YAML 1.1 reads dotless scientific notation ("1e-07") as a
+            # STRING; coerce to the field's numeric type so e.g.
+            # lr=1e-07 does not silently poison the optimizer
+            cur = data[k]
+            if isinstance(parsed, str) and isinstance(cur, (int, float)) \
+                    and not isinstance(cur, bool):
+                try:
+                    parsed = type(cur)(float(parsed)) \
+                        if isinstance(cur, int) else float(parsed)
+                except ValueError:
+                    pass
+            data[k] = parsed
         return Config.from_dict(data)

@@ -31,7 +31,10 @@ def predict_flow(model, img1_raw, img2_raw, mean_bgr, flow_scale_finest,
     """Run the model and apply the eval post-processing; returns [B,2,H,W]."""
     x1 = preprocess_images(img1_raw.float(), mean_bgr)
     x2 = preprocess_images(img2_raw.float(), mean_bgr)
-    out = model(torch.cat([x1, x2], dim=1))
+    x = torch.cat([x1, x2], dim=1)
+    if x.is_cuda:  # NHWC so eval hits the MFMA kernels, not a fallback
+        x = x.contiguous(memory_format=torch.channels_last)
+    out = model(x)
     flows = out[0] if isinstance(out, tuple) else out  # joint models
     mult, cmin, cmax = EVAL_POSTPROC.get(dataset, EVAL_POSTPROC["flying_chairs"])
     pred = flows[0].float() * flow_scale_finest * mult
