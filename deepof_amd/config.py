@@ -60,6 +60,7 @@ class Config:
     weight_decay: float = 0.0
     max_epochs: int = 110
     grad_clip: float = 0.0
+    grad_accumulation: int = 1           # micro-batches per optimizer step
 
     # runtime
     precision: str = "bf16"              # bf16|fp32

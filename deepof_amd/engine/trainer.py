@@ -107,6 +107,7 @@ class Trainer:
         )
         self.epoch = 0
         self.global_step = 0
+        self._micro_step = 0
 
         self.run_dir = os.path.join(cfg.log_dir, cfg.run_name)
         if self.rank == 0:
@@ -255,23 +256,34 @@ class Trainer:
                 parts["action_acc"] = float(
                     (logits.argmax(1) == labels).float().mean())
 
-        total.backward()
+        # gradient accumulation: micro-batches sum into .grad (and the
+        # DDP flat buckets); the all-reduce and optimizer step run only
+        # on the boundary micro-batch
+        accum = max(1, cfg.grad_accumulation)
+        boundary = (self._micro_step + 1) % accum == 0
+        if accum > 1:
+            total = total / accum
         if isinstance(self.model, BucketedDataParallel):
-            self.model.finish_gradient_sync()
-        if cfg.grad_clip > 0:
-            torch.nn.utils.clip_grad_norm_(self.model.parameters(),
-                                           cfg.grad_clip)
+            self.model.accumulate_only = not boundary
+        total.backward()
+        self._micro_step += 1
 
-        loss_val = float(total.detach())
+        loss_val = float(total.detach()) * accum
         if math.isnan(loss_val) or math.isinf(loss_val):
             raise FloatingPointError(
                 f"Model diverged (loss={loss_val}) at step {self.global_step}"
             )
 
-        for g in self.optimizer.param_groups:
-            g["lr"] = self.current_lr()
-        self.optimizer.step()
-        self._zero_grads()
+        if boundary:
+            if isinstance(self.model, BucketedDataParallel):
+                self.model.finish_gradient_sync()
+            if cfg.grad_clip > 0:
+                torch.nn.utils.clip_grad_norm_(self.model.parameters(),
+                                               cfg.grad_clip)
+            for g in self.optimizer.param_groups:
+                g["lr"] = self.current_lr()
+            self.optimizer.step()
+            self._zero_grads()
         self.global_step += 1
         parts["total"] = loss_val
         return parts
@@ -329,6 +341,9 @@ class Trainer:
                             or not self.try_resume()):
                         raise
                     self._zero_grads()
+                    self._micro_step = 0  # restart accumulation cleanly
+                    if isinstance(self.model, BucketedDataParallel):
+                        self.model.accumulate_only = False
                     print(f"[deepof] {e}; restarted from checkpoint "
                           f"({nan_restarts}/{cfg.nan_restart_limit})")
                     break

@@ -68,6 +68,10 @@ class BucketedDataParallel(torch.nn.Module):
         super().__init__()
         self.module = module
         self.pg = process_group
+        # gradient-accumulation support: while True, backward passes
+        # accumulate into the flat buckets WITHOUT all-reducing (the
+        # boundary micro-batch clears it and reduces the sums)
+        self.accumulate_only = False
         self.world = dist.get_world_size(process_group) if is_distributed() else 1
         self._hooks = []
         self._buckets: list[_Bucket] = []
@@ -142,8 +146,11 @@ class BucketedDataParallel(torch.nn.Module):
         b = self._param_bucket[id(param)]
         b.pending -= 1
         if b.pending == 0:
-            b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
-                                     group=self.pg, async_op=True)
+            if self.accumulate_only:
+                b.pending = len(b.params)  # re-arm, no reduce this pass
+            else:
+                b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
+                                         group=self.pg, async_op=True)
 
     def forward(self, *args, **kwargs):
         return self.module(*args, **kwargs)
