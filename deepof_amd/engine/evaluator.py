@@ -129,6 +129,8 @@ def evaluate_accuracy(model, loader, mean_bgr, device,
         labels = batch["label"].to(device, non_blocking=True)
         x = torch.cat([preprocess_images(img1.float(), mean_bgr),
                        preprocess_images(img2.float(), mean_bgr)], dim=1)
+        if x.is_cuda:
+            x = x.contiguous(memory_format=torch.channels_last)
         out = model(x)
         logits = out[1] if isinstance(out, tuple) else out
         correct += int((logits.argmax(1) == labels).sum())

@@ -144,6 +144,33 @@ class FlowDecoder(nn.Module):
                 concat_ch = (raw + 63) // 64 * 64
                 self._concat_pad[i + 1] = concat_ch - raw
 
+    def _pad_zeros(self, skip, idx, h, w):
+        """Cached read-only zero pad for the channel-64 alignment.
+
+        The zeros MUST match the memory format of the other cat inputs:
+        one NCHW tensor in a channels_last cat demotes the output to
+        NCHW and every consumer then pays a layout-transposing copy
+        (measured +10 ms/step on the r02 bench).  Cached per shape —
+        the tensor is only ever READ (cat copies it), so reuse across
+        steps is safe and removes a per-step alloc+fill.
+        """
+        fmt = (torch.channels_last
+               if skip.is_contiguous(memory_format=torch.channels_last)
+               else torch.contiguous_format)
+        key = (idx, skip.shape[0], h, w, skip.dtype, str(skip.device), fmt)
+        cache = getattr(self, "_pad_cache", None)
+        if cache is None:
+            cache = self._pad_cache = {}
+        z = cache.get(key)
+        if z is None:
+            if len(cache) > 16:  # bound memory across batch-size changes
+                cache.clear()
+            z = torch.empty((skip.shape[0], self._concat_pad[idx], h, w),
+                            dtype=skip.dtype, device=skip.device,
+                            memory_format=fmt).zero_()
+            cache[key] = z
+        return z
+
     def forward(self, features: list[torch.Tensor]) -> list[torch.Tensor]:
         assert len(features) == self.num_scales
         flows = []
@@ -159,17 +186,6 @@ class FlowDecoder(nn.Module):
                 h, w = skip.shape[-2:]
                 parts = [skip, up_feat[..., :h, :w], up_flow[..., :h, :w]]
                 if self._concat_pad[i + 1]:
-                    # zeros MUST match the memory format of the other
-                    # inputs: one NCHW tensor in a channels_last cat
-                    # demotes the output to NCHW and every consumer then
-                    # pays a layout-transposing copy (measured +10 ms/
-                    # step on the r02 bench before this guard)
-                    fmt = (torch.channels_last if skip.is_contiguous(
-                        memory_format=torch.channels_last)
-                        else torch.contiguous_format)
-                    parts.append(torch.empty(
-                        (skip.shape[0], self._concat_pad[i + 1], h, w),
-                        dtype=skip.dtype, device=skip.device,
-                        memory_format=fmt).zero_())
+                    parts.append(self._pad_zeros(skip, i + 1, h, w))
                 x = torch.cat(parts, dim=1)
         return flows
