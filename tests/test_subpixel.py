@@ -92,3 +92,46 @@ def test_fused_deconv_module_cpu_fallback():
     flows = dec(feats)
     assert flows[0].shape == (1, 2, 4, 6)
     assert flows[1].shape == (1, 2, 8, 12)
+
+
+def test_fused_deconv_checkpoint_roundtrip(tmp_path):
+    """FusedDeconvAct params live in the wrapped ConvTranspose2d:
+    state_dict round-trips and bilinear init survives re-init."""
+    import torch
+
+    from deepof_amd.models import FlowNetS
+
+    torch.manual_seed(0)
+    m1 = FlowNetS()
+    path = tmp_path / "m.pt"
+    torch.save(m1.state_dict(), path)
+    m2 = FlowNetS()
+    missing, unexpected = m2.load_state_dict(
+        torch.load(path, weights_only=True))
+    assert not missing and not unexpected
+    x = torch.randn(1, 6, 64, 96)
+    f1 = m1(x)
+    f2 = m2(x)
+    for a, b in zip(f1, f2):
+        assert torch.equal(a, b)
+
+
+def test_decoder_pad_cache_reuse():
+    """The cached pad-zeros tensor is reused across forwards (same
+    object), stays zero, and respects batch-size changes."""
+    import torch
+
+    from deepof_amd.models.common import FlowDecoder
+
+    dec = FlowDecoder([64, 32], [16], act="elu", flow_channels=2)
+    assert dec._concat_pad[1] == 64 - (32 + 16 + 2)
+    feats = [torch.randn(2, 64, 4, 6), torch.randn(2, 32, 8, 12)]
+    dec(feats)
+    (key1, z1), = dec._pad_cache.items()
+    dec(feats)
+    (key2, z2), = dec._pad_cache.items()
+    assert key1 == key2 and z1 is z2
+    assert z1.abs().sum() == 0
+    feats4 = [torch.randn(4, 64, 4, 6), torch.randn(4, 32, 8, 12)]
+    dec(feats4)
+    assert len(dec._pad_cache) == 2  # both batch sizes cached
