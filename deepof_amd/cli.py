@@ -70,7 +70,10 @@ def main(argv=None):
                                   drop_last=False, num_workers=2)
         mean = DATASET_MEANS.get(cfg.dataset, (127.5, 127.5, 127.5))
         aee = evaluate_aee(model, loader, mean, flow_scales[0], device,
-                           cfg.dataset, dump_dir=args.dump_dir)
+                           cfg.dataset, dump_dir=args.dump_dir,
+                           mult=cfg.eval_mult,
+                           clip=tuple(cfg.eval_clip) if cfg.eval_clip
+                           else None)
         print(f"AEE: {aee:.4f}")
     elif args.command == "infer":
         import numpy as np

@@ -74,6 +74,13 @@ class Config:
     nan_restart_limit: int = 3           # auto-restarts from ckpt on NaN
     profile_steps: int = 0               # torch.profiler trace of N steps
 
+    # eval post-processing overrides (None = dataset defaults; the
+    # reference varies these per config: chairs pr1 x2 clip [-300,250],
+    # chairs-VGG clip [-204.479, 201.3478], Sintel x amplifier 3 —
+    # SURVEY §2.5 eval row)
+    eval_mult: Optional[float] = None
+    eval_clip: Optional[tuple] = None    # (min, max)
+
     # action head (UCF101 joint training)
     action_classes: int = 0              # >0 enables the action head
     action_weight: float = 1.0

@@ -24,10 +24,24 @@ EVAL_POSTPROC = {
 }
 
 
+def _postproc(dataset, mult=None, clip=None):
+    """Dataset defaults with optional per-config overrides (the
+    reference varies mult/clip per model variant)."""
+    d_mult, d_min, d_max = EVAL_POSTPROC.get(
+        dataset, EVAL_POSTPROC["flying_chairs"])
+    if mult is not None:
+        d_mult = mult
+    if clip is not None:
+        d_min, d_max = clip
+    return d_mult, d_min, d_max
+
+
 @torch.no_grad()
 def predict_flow(model, img1_raw, img2_raw, mean_bgr, flow_scale_finest,
                  dataset: str = "flying_chairs",
-                 gt_size: tuple[int, int] | None = None) -> torch.Tensor:
+                 gt_size: tuple[int, int] | None = None,
+                 mult: float | None = None,
+                 clip: tuple | None = None) -> torch.Tensor:
     """Run the model and apply the eval post-processing; returns [B,2,H,W]."""
     x1 = preprocess_images(img1_raw.float(), mean_bgr)
     x2 = preprocess_images(img2_raw.float(), mean_bgr)
@@ -36,7 +50,7 @@ def predict_flow(model, img1_raw, img2_raw, mean_bgr, flow_scale_finest,
         x = x.contiguous(memory_format=torch.channels_last)
     out = model(x)
     flows = out[0] if isinstance(out, tuple) else out  # joint models
-    mult, cmin, cmax = EVAL_POSTPROC.get(dataset, EVAL_POSTPROC["flying_chairs"])
+    mult, cmin, cmax = _postproc(dataset, mult, clip)
     pred = flows[0].float() * flow_scale_finest * mult
     pred = pred.clamp(cmin, cmax)
     if gt_size is not None and tuple(pred.shape[-2:]) != tuple(gt_size):
@@ -48,7 +62,9 @@ def predict_flow(model, img1_raw, img2_raw, mean_bgr, flow_scale_finest,
 @torch.no_grad()
 def evaluate_aee(model, loader, mean_bgr, flow_scale_finest, device,
                  dataset: str = "flying_chairs", max_batches=None,
-                 dump_dir: str | None = None, dump_every: int = 10) -> float:
+                 dump_dir: str | None = None, dump_every: int = 10,
+                 mult: float | None = None,
+                 clip: tuple | None = None) -> float:
     """AEE over a loader; optionally dump flow color maps, warped
     frames and predicted .flo files (parity with the reference's eval
     artifacts, flyingChairsTrain.py:272-291)."""
@@ -68,9 +84,8 @@ def evaluate_aee(model, loader, mean_bgr, flow_scale_finest, device,
             x = (vol.float() - mean.repeat(T).view(1, -1, 1, 1)) / 255.0
             out = model(x)
             flows = out[0] if isinstance(out, tuple) else out
-            mult, cmin, cmax = EVAL_POSTPROC.get(
-                dataset, EVAL_POSTPROC["flying_chairs"])
-            pred = (flows[0][:, :2].float() * flow_scale_finest * mult
+            m_, cmin, cmax = _postproc(dataset, mult, clip)
+            pred = (flows[0][:, :2].float() * flow_scale_finest * m_
                     ).clamp(cmin, cmax)
             gt = gt[:, :2]
             if tuple(pred.shape[-2:]) != tuple(gt.shape[-2:]):
@@ -80,7 +95,8 @@ def evaluate_aee(model, loader, mean_bgr, flow_scale_finest, device,
             img2 = batch["img2"].to(device, non_blocking=True)
             pred = predict_flow(model, img1, img2, mean_bgr,
                                 flow_scale_finest, dataset,
-                                gt_size=tuple(gt.shape[-2:]))
+                                gt_size=tuple(gt.shape[-2:]),
+                                mult=mult, clip=clip)
         total += float(ops.endpoint_error_sum(pred, gt))
         count += gt.shape[0] * gt.shape[-2] * gt.shape[-1]
         if dump_dir is not None and i % dump_every == 0:

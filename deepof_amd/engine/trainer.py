@@ -378,6 +378,8 @@ class Trainer:
                     aee = evaluate_aee(
                         self.raw_model, val_loader, self.mean_bgr,
                         self.flow_scales[0], self.device, cfg.dataset,
+                        mult=cfg.eval_mult,
+                        clip=tuple(cfg.eval_clip) if cfg.eval_clip else None,
                     )
                     self.log_metrics({"epoch": self.epoch, "aee": aee})
                     print(f"[deepof] ***Test: epoch {self.epoch} "

@@ -275,3 +275,28 @@ def test_evaluate_aee_volume():
     aee = evaluate_aee(model, dl, (70.0, 83.0, 92.0), scales[0], "cpu",
                        "sintel")
     assert aee > 0
+
+
+def test_eval_postproc_overrides():
+    """Per-config eval overrides (the VGG chairs variant
+    clips to [-204.479, 201.3478]; version1 sintel uses a different
+    amplifier) flow through predict_flow."""
+    import torch
+
+    from deepof_amd.engine.evaluator import _postproc, predict_flow
+    from deepof_amd.models import build_model
+
+    assert _postproc("flying_chairs") == (2.0, -300.0, 250.0)
+    assert _postproc("flying_chairs", clip=(-204.479, 201.3478)) == \
+        (2.0, -204.479, 201.3478)
+    assert _postproc("sintel", mult=100.0) == (100.0, -420.621, 426.311)
+
+    torch.manual_seed(0)
+    model, scales, _ = build_model("flownets")
+    img = torch.rand(1, 3, 64, 96) * 255
+    p_default = predict_flow(model, img, img, (127.5,) * 3, scales[0],
+                             "flying_chairs")
+    p_clip = predict_flow(model, img, img, (127.5,) * 3, scales[0],
+                          "flying_chairs", clip=(-0.01, 0.01))
+    assert p_clip.abs().max() <= 0.01
+    assert p_default.shape == p_clip.shape
