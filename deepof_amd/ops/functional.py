@@ -94,7 +94,10 @@ def lrn(x, depth_radius: int = 4, bias: float = 1.0, alpha: float = 1.0,
         beta: float = 0.7) -> torch.Tensor:
     if x.is_cuda:
         if x.requires_grad and torch.is_grad_enabled():
-            raise RuntimeError("lrn HIP path has no backward; detach the input")
+            # differentiable torch-op path (a future loss backprops
+            # through the image branch); the HIP kernel serves the
+            # no-grad image-pyramid hot path
+            return ref.lrn(x, depth_radius, bias, alpha, beta)
         return require_hip().lrn_forward(x.contiguous(), depth_radius, bias, alpha, beta)
     return ref.lrn(x, depth_radius, bias, alpha, beta)
 
