@@ -108,6 +108,7 @@ class Trainer:
         self.epoch = 0
         self.global_step = 0
         self._micro_step = 0
+        self.best_aee = float("inf")
 
         self.run_dir = os.path.join(cfg.log_dir, cfg.run_name)
         if self.rank == 0:
@@ -384,6 +385,17 @@ class Trainer:
                     self.log_metrics({"epoch": self.epoch, "aee": aee})
                     print(f"[deepof] ***Test: epoch {self.epoch} "
                           f"AEE {aee:.4f}")
+                    if aee < self.best_aee:
+                        self.best_aee = aee
+                        state = {
+                            "model": self.raw_model.state_dict(),
+                            "epoch": self.epoch, "aee": aee,
+                            "config": self.cfg.to_dict(),
+                        }
+                        tmp = os.path.join(self.run_dir, "ckpt_best.pt.tmp")
+                        torch.save(state, tmp)
+                        os.replace(tmp, os.path.join(self.run_dir,
+                                                     "ckpt_best.pt"))
                 if "label" in sample and cfg.action_classes > 0:
                     from .evaluator import evaluate_accuracy
 

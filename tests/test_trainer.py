@@ -111,3 +111,43 @@ def test_grad_accumulation_equivalence(tmp_path):
     # batch; Adam's rsqrt amplifies that on near-zero second moments
     for pa, pb in zip(t_acc.model.parameters(), t_one.model.parameters()):
         torch.testing.assert_close(pa, pb, rtol=1e-3, atol=1e-4)
+
+
+def test_best_checkpoint_saved(tmp_path):
+    """ckpt_best.pt tracks the lowest eval AEE (model-only, with the
+    achieving epoch recorded)."""
+    import os
+
+    import torch
+
+    from deepof_amd.config import Config
+    from deepof_amd.engine import Trainer
+
+    cfg = Config.from_dict(dict(
+        dataset="synthetic", image_size=(48, 64), batch_size=2,
+        num_workers=0, model="flownets", precision="fp32", device="cpu",
+        log_dir=str(tmp_path), run_name="best", eval_interval_epochs=1,
+        max_epochs=1, resume=False,
+    ))
+    tr = Trainer(cfg)
+    # shrink the epoch: 4 samples -> 2 steps
+    from deepof_amd.data import SyntheticFlowDataset
+
+    import deepof_amd.engine.trainer as T
+
+    orig = T.build_datasets
+
+    def tiny(cfg_):
+        return (SyntheticFlowDataset(4, 48, 64),
+                SyntheticFlowDataset(4, 48, 64, seed=1))
+
+    T.build_datasets = tiny
+    try:
+        tr.fit(max_epochs=1)
+    finally:
+        T.build_datasets = orig
+    best = os.path.join(str(tmp_path), "best", "ckpt_best.pt")
+    assert os.path.exists(best)
+    state = torch.load(best, weights_only=False)
+    assert state["aee"] == tr.best_aee
+    assert "model" in state and state["epoch"] == 1
