@@ -31,6 +31,14 @@ def main(argv=None):
                         help="write flow color maps / .flo / warped frames")
     p_eval.add_argument("overrides", nargs="*")
 
+    p_serve = sub.add_parser("serve",
+                             help="HTTP flow inference service (FastAPI)")
+    p_serve.add_argument("--config", type=str, default=None)
+    p_serve.add_argument("--checkpoint", type=str, default=None)
+    p_serve.add_argument("--host", type=str, default="127.0.0.1")
+    p_serve.add_argument("--port", type=int, default=8000)
+    p_serve.add_argument("overrides", nargs="*")
+
     p_infer = sub.add_parser("infer", help="predict flow for an image pair")
     p_infer.add_argument("--config", type=str, default=None)
     p_infer.add_argument("--checkpoint", type=str, required=True)
@@ -48,6 +56,10 @@ def main(argv=None):
         from .engine import Trainer
 
         Trainer(cfg).fit(max_steps=args.max_steps)
+    elif args.command == "serve":
+        from .serve import serve_from_config
+
+        serve_from_config(cfg, args.checkpoint, args.host, args.port)
     elif args.command == "eval":
         import torch
 
