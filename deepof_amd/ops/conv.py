@@ -19,6 +19,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 _ACT_CODE = {None: 0, "none": 0, "elu": 1, "leaky_relu": 2, "relu": 3}
+_CODE_ACT = {0: "none", 1: "elu", 2: "leaky_relu", 3: "relu"}
 _dispatch_cache: dict[tuple, str] = {}
 _wrw_cache: dict[tuple, str] = {}
 
@@ -60,7 +61,7 @@ class _MiopenConvFn(torch.autograd.Function):
         y = torch.ops.aten.convolution(
             x, w, bias, [stride, stride], [pad, pad], [1, 1], False,
             [0, 0], 1)
-        y = _act(y, {v: k for k, v in _ACT_CODE.items()}[act_code])
+        y = _act(y, _CODE_ACT[act_code])
         y = y.contiguous(memory_format=torch.channels_last)
         ctx.save_for_backward(x, w, y)
         ctx.meta = (stride, pad, act_code, bias is not None)
