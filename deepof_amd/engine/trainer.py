@@ -180,7 +180,11 @@ class Trainer:
                 set_to_none=self.device.type != "cuda")
 
     # -- the step ---------------------------------------------------------
-    def train_step(self, batch) -> dict:
+    def train_step(self, batch, log_scales: bool = False) -> dict:
+        """One micro-batch.  log_scales=True additionally reports the
+        per-scale loss components (the reference's display-interval
+        print, flyingChairsTrain.py:183-201) — off by default since
+        each component read is a device sync."""
         cfg = self.cfg
         from ..losses.unsup import preprocess_images
 
@@ -205,6 +209,9 @@ class Trainer:
             res = self._mf_loss(flows, vol)
             total = res["total"]
             parts["unsup"] = float(total.detach())
+            if log_scales:
+                parts["scale_losses"] = [
+                    round(float(s.detach()), 6) for s in res["scales"]]
         else:
             img1 = batch["img1"].to(self.device, non_blocking=True)
             img2 = batch["img2"].to(self.device, non_blocking=True)
@@ -230,6 +237,10 @@ class Trainer:
             res = self.unsup_loss(flows, geo1, geo2)
             total = cfg.photo_weight * res["total"]
             parts["unsup"] = float(res["total"].detach())
+            if log_scales:
+                parts["scale_losses"] = [
+                    round(float(s["total"].detach()), 6)
+                    for s in res["scales"]]
             if self.guided_loss is not None and "flow" in batch:
                 gt = batch["flow"].to(self.device, non_blocking=True)
                 g = self.guided_loss(flows, gt)
@@ -327,7 +338,10 @@ class Trainer:
                 batches = CudaPrefetcher(loader, self.device)
             for i, batch in enumerate(batches):
                 try:
-                    parts = self.train_step(batch)
+                    parts = self.train_step(
+                        batch,
+                        log_scales=(self.global_step + 1)
+                        % cfg.log_interval == 0)
                 except FloatingPointError as e:
                     # divergence guard (reference asserts and dies,
                     # flyingChairsTrain.py:203); here: restart from the

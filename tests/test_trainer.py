@@ -183,3 +183,23 @@ def test_every_model_family_trains(tmp_path, model):
     changed = any(not torch.equal(a, b) for a, b in
                   zip(before, tr.model.parameters()))
     assert changed, f"{model}: no parameter moved after a step"
+
+
+def test_per_scale_loss_logging(tmp_path):
+    from deepof_amd.config import Config
+    from deepof_amd.data import SyntheticFlowDataset
+    from deepof_amd.engine import Trainer
+
+    cfg = Config.from_dict(dict(
+        dataset="synthetic", image_size=(48, 64), batch_size=2,
+        num_workers=0, model="flownets", precision="fp32", device="cpu",
+        log_dir=str(tmp_path), run_name="ps", resume=False,
+    ))
+    tr = Trainer(cfg)
+    ds = SyntheticFlowDataset(2, 48, 64)
+    batch = {k: v.unsqueeze(0) for k, v in ds[0].items()}
+    parts = tr.train_step(batch, log_scales=True)
+    assert len(parts["scale_losses"]) == 6  # one per pyramid scale
+    assert all(s >= 0 for s in parts["scale_losses"])
+    parts2 = tr.train_step(batch)
+    assert "scale_losses" not in parts2
