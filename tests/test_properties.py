@@ -2,6 +2,7 @@
 invariants, sub-pixel plan tables, config override round-trip."""
 
 import numpy as np
+import pytest
 import torch
 from hypothesis import given, settings
 from hypothesis import strategies as st
@@ -87,3 +88,73 @@ def test_config_override_roundtrip(lr, bs, lam):
     assert cfg.lambda_smooth == lam
     # round-trip through dict preserves everything
     assert Config.from_dict(cfg.to_dict()) == cfg
+
+
+@settings(max_examples=15, deadline=None)
+@given(seed=st.integers(0, 2**31 - 1))
+def test_photometric_augment_range(seed):
+    """Photometric augmentation must keep images in [0, 255] and apply
+    the SAME transform to both frames of a pair (the loss warps one
+    onto the other)."""
+    from deepof_amd.utils.augment import photometric_augment
+
+    g = torch.Generator().manual_seed(seed)
+    img1 = torch.rand(2, 3, 16, 20, generator=g) * 255
+    img2 = torch.rand(2, 3, 16, 20, generator=g) * 255
+    a1, a2 = photometric_augment(img1, img2, generator=g)
+    for a in (a1, a2):
+        assert a.shape == img1.shape
+        assert float(a.min()) >= 0.0 and float(a.max()) <= 255.0
+    # identical inputs -> identical outputs up to the additive noise
+    b1, b2 = photometric_augment(img1, img1,
+                                 generator=torch.Generator().manual_seed(7),
+                                 noise_sigma=0.0)
+    assert torch.allclose(b1, b2)
+
+
+@settings(max_examples=15, deadline=None)
+@given(seed=st.integers(0, 2**31 - 1))
+def test_geometric_augment_shape_and_shared_transform(seed):
+    from deepof_amd.utils.augment import geometric_augment
+
+    g = torch.Generator().manual_seed(seed)
+    img = torch.rand(2, 3, 16, 20, generator=g) * 255
+    a1, a2 = geometric_augment(img, img,
+                               generator=torch.Generator().manual_seed(3))
+    assert a1.shape == img.shape
+    # same input + same transform -> both outputs identical
+    assert torch.allclose(a1, a2)
+
+
+def test_fused_adam_cpu_matches_torch_adam():
+    """The CPU foreach path (the numerics reference for the HIP kernel)
+    must match torch.optim.Adam exactly (no weight decay: torch's Adam
+    uses decoupled grad add like ours)."""
+    from deepof_amd.engine.optim import FusedAdam
+
+    torch.manual_seed(0)
+    shapes = [(64,), (8, 8), (3, 4, 5)]
+    pa = [torch.randn(s, requires_grad=True) for s in shapes]
+    pb = [p.detach().clone().requires_grad_(True) for p in pa]
+    grads = [torch.randn(s) for s in shapes]
+    for p, g in zip(pa, grads):
+        p.grad = g.clone()
+    for p, g in zip(pb, grads):
+        p.grad = g.clone()
+    oa = FusedAdam(pa, lr=1e-2, betas=(0.9, 0.999), eps=1e-8)
+    ob = torch.optim.Adam(pb, lr=1e-2, betas=(0.9, 0.999), eps=1e-8)
+    for _ in range(5):
+        oa.step()
+        ob.step()
+    for a, b in zip(pa, pb):
+        torch.testing.assert_close(a, b, rtol=1e-6, atol=1e-8)
+
+
+def test_read_flo_rejects_bad_magic(tmp_path_factory):
+    from deepof_amd.utils import read_flo
+
+    p = tmp_path_factory.mktemp("bad") / "x.flo"
+    p.write_bytes(b"\x00\x00\x00\x00" + b"\x01\x00\x00\x00" * 2 + b"\x00" * 8)
+    with pytest.raises(Exception):
+        read_flo(p)
+
