@@ -203,3 +203,28 @@ def test_per_scale_loss_logging(tmp_path):
     assert all(s >= 0 for s in parts["scale_losses"])
     parts2 = tr.train_step(batch)
     assert "scale_losses" not in parts2
+
+
+def test_cpu_training_loss_decreases(tmp_path):
+    """30 optimizer steps on one repeated synthetic batch must reduce
+    the unsupervised loss (CPU counterpart of the GPU convergence
+    smoke) — catches sign/scale regressions in the loss/optimizer
+    wiring without a GPU."""
+    import numpy as np
+
+    from deepof_amd.config import Config
+    from deepof_amd.data import SyntheticFlowDataset
+    from deepof_amd.engine import Trainer
+
+    cfg = Config.from_dict(dict(
+        dataset="synthetic", image_size=(32, 48), batch_size=2,
+        num_workers=0, model="flownets", precision="fp32", device="cpu",
+        log_dir=str(tmp_path), run_name="conv", lr=1e-4, resume=False,
+    ))
+    tr = Trainer(cfg)
+    ds = SyntheticFlowDataset(2, 32, 48)
+    batch = {k: torch.stack([ds[0][k], ds[1][k]]) for k in ds[0]}
+    losses = [tr.train_step(batch)["total"] for _ in range(30)]
+    assert all(np.isfinite(l) for l in losses)
+    assert min(losses[-5:]) < losses[0], (losses[0], losses[-5:])
+
