@@ -59,6 +59,7 @@ def create_app(model, flow_scales, mean_bgr, device,
                dataset: str = "flying_chairs", precision: str = "bf16",
                eval_mult=None, eval_clip=None):
     import base64
+    import time
 
     from fastapi import Body, FastAPI
     from fastapi.responses import JSONResponse, Response
@@ -69,6 +70,24 @@ def create_app(model, flow_scales, mean_bgr, device,
     model.eval()
     use_bf16 = precision == "bf16" and device.type == "cuda"
 
+    # Prometheus metrics (own registry so repeated create_app calls in
+    # one process don't collide)
+    from prometheus_client import (CONTENT_TYPE_LATEST, CollectorRegistry,
+                                   Counter, Histogram, generate_latest)
+
+    registry = CollectorRegistry()
+    req_count = Counter("deepof_flow_requests_total",
+                        "flow requests served", ["format"],
+                        registry=registry)
+    req_latency = Histogram("deepof_flow_latency_seconds",
+                            "end-to-end /flow latency",
+                            registry=registry)
+
+    @app.get("/metrics")
+    def metrics():
+        return Response(content=generate_latest(registry),
+                        media_type=CONTENT_TYPE_LATEST)
+
     @app.get("/healthz")
     def healthz():
         return {"status": "ok", "model": type(model).__name__,
@@ -76,6 +95,8 @@ def create_app(model, flow_scales, mean_bgr, device,
 
     @app.post("/flow")
     def flow(req: FlowRequest = Body(...), format: str = "json"):
+        t_start = time.perf_counter()
+        req_count.labels(format=format).inc()
         t1 = _decode_image(base64.b64decode(req.img1)).to(device)
         t2 = _decode_image(base64.b64decode(req.img2),
                            size=t1.shape[-2:]).to(device)
@@ -85,6 +106,7 @@ def create_app(model, flow_scales, mean_bgr, device,
                                 dataset, gt_size=tuple(t1.shape[-2:]),
                                 mult=eval_mult, clip=eval_clip)
         f = pred[0].float().permute(1, 2, 0).cpu().numpy()
+        req_latency.observe(time.perf_counter() - t_start)
         if format == "flo":
             return Response(content=_flo_bytes(f),
                             media_type="application/octet-stream")

@@ -73,3 +73,12 @@ def test_flow_png(client):
     assert r.status_code == 200
     img = Image.open(io.BytesIO(r.content))
     assert img.size == (96, 64)
+
+
+def test_metrics_endpoint(client):
+    client.post("/flow", json=_payload())
+    r = client.get("/metrics")
+    assert r.status_code == 200
+    body = r.text
+    assert "deepof_flow_requests_total" in body
+    assert "deepof_flow_latency_seconds" in body
