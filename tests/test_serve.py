@@ -82,3 +82,15 @@ def test_metrics_endpoint(client):
     body = r.text
     assert "deepof_flow_requests_total" in body
     assert "deepof_flow_latency_seconds" in body
+
+
+def test_flow_odd_image_size(client):
+    """Arbitrary (non-power-of-two) image sizes must round-trip: the
+    decoder crops 2x-upsampled skips to odd sizes."""
+    import base64
+
+    payload = {"img1": base64.b64encode(_img_bytes(5, h=70, w=90)).decode(),
+               "img2": base64.b64encode(_img_bytes(6, h=70, w=90)).decode()}
+    r = client.post("/flow", json=payload)
+    assert r.status_code == 200
+    assert r.json()["shape"] == [70, 90, 2]
