@@ -114,25 +114,14 @@ def conv_transpose2d_subpixel(x: torch.Tensor, w: torch.Tensor,
             (B, N, out_h, out_w), device=x.device, dtype=torch.bfloat16
         ).contiguous(memory_format=torch.channels_last)
     b = bias if bias is not None else _empty().to(x.device)
-    if M % 64 == 0:
-        # fast path: 1 pack launch + 1 all-parity conv launch
-        wcl = w.contiguous(memory_format=torch.channels_last)
-        pack_tab, conv_tab = _plan_tabs(R, S, pad, N, M, x.device)
-        wp = hip.subpixel_pack(wcl, pack_tab, R, S)
-        hip.conv2d_fwd_subpixel4(x, wp, b, out, conv_tab, N, act, 2,
-                                 out_coff)
-        return out
-    # generic path: 4 separate parity launches with torch-gathered
-    # sub-filters (odd channel counts)
-    w_nm = w.transpose(0, 1)
-    for uy, taps_y, pad_y in _axis_plan(R, pad):
-        for ux, taps_x, pad_x in _axis_plan(S, pad):
-            if not taps_y or not taps_x or uy >= out_h or ux >= out_w:
-                continue
-            sub = (w_nm[:, :, taps_y][:, :, :, taps_x]
-                   .contiguous(memory_format=torch.channels_last))
-            hip.conv2d_fwd_strided(x, sub, b, out, pad_y, pad_x, act,
-                                   2, uy, ux, out_coff)
+    # every dispatcher pre-gates on subpixel_eligible (M % 64 == 0, the
+    # kernel's reduction-tile requirement), so the single-launch path is
+    # the only one: 1 pack launch + 1 all-parity conv launch
+    assert M % 64 == 0, "gate with subpixel_eligible() first"
+    wcl = w.contiguous(memory_format=torch.channels_last)
+    pack_tab, conv_tab = _plan_tabs(R, S, pad, N, M, x.device)
+    wp = hip.subpixel_pack(wcl, pack_tab, R, S)
+    hip.conv2d_fwd_subpixel4(x, wp, b, out, conv_tab, N, act, 2, out_coff)
     return out
 
 

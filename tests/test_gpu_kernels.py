@@ -410,3 +410,29 @@ def test_fused_deconv_module_gpu():
                                rtol=0.08, atol=0.02)
     torch.testing.assert_close(m.deconv.weight.grad.float(),
                                m2.weight.grad.float(), rtol=0.08, atol=0.02)
+
+
+def test_conv2d_fwd_strided_single_parity():
+    """The single-parity strided-out primitive (the building block the
+    all-parity launch composes; also the r03 concat-epilogue writer):
+    4 explicit parity calls == ConvTranspose2d."""
+    from deepof_amd.ops.deconv import _axis_plan
+
+    torch.manual_seed(3)
+    B, C, K, H, W = 2, 16, 32, 10, 12
+    x = (torch.randn(B, C, H, W, device=DEV).bfloat16()
+         .contiguous(memory_format=torch.channels_last))
+    w_ct = torch.randn(C, K, 4, 4, device=DEV).bfloat16() * 0.1
+    out = torch.empty((B, K, 2 * H, 2 * W), device=DEV,
+                      dtype=torch.bfloat16).contiguous(
+        memory_format=torch.channels_last)
+    w_nm = w_ct.transpose(0, 1)
+    for uy, ty, py in _axis_plan(4, 1):
+        for ux, tx, px in _axis_plan(4, 1):
+            sub = (w_nm[:, :, ty][:, :, :, tx]
+                   .contiguous(memory_format=torch.channels_last))
+            _hip().conv2d_fwd_strided(x, sub, torch.Tensor(), out,
+                                      py, px, 0, 2, uy, ux, 0)
+    want = torch.nn.functional.conv_transpose2d(
+        x.float(), w_ct.float(), stride=2, padding=1)
+    torch.testing.assert_close(out.float(), want, rtol=0.06, atol=0.06)
